@@ -1,0 +1,88 @@
+"""Fused linear + cross-entropy over the 50368 vocab.
+
+Replaces the reference's flash_attn.ops.cross_entropy fused CE
+(SURVEY.md §2.3). MI355X-native design: the [N, d] @ [V, d]^T logits GEMM
+runs on hipBLASLt (torch.matmul) in row chunks; a HIP kernel computes the
+log-sum-exp + NLL and writes the softmax gradient IN PLACE into the logits
+chunk, so the full [N, 50368] logits tensor is never materialized and the
+backward GEMMs (dH = dL @ W, dW = dL^T @ H) consume the in-place gradient.
+This keeps HBM traffic at O(chunk * V) instead of O(N * V) live memory —
+sized for 288 GB HBM3E but dominated by bandwidth, not capacity.
+
+CPU fallback: the same chunked math in plain PyTorch (fp32 log_softmax).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from . import use_hip, hip_ext
+
+_CHUNK = 4096  # rows per logits chunk
+
+
+class _FusedLinearCrossEntropy(torch.autograd.Function):
+    """loss = mean_i CE(h_i @ W^T, t_i) with grads for h and W, chunked."""
+
+    @staticmethod
+    def forward(ctx, hidden, weight, targets, use_kernel: bool):
+        N, D = hidden.shape
+        V = weight.shape[0]
+        device = hidden.device
+        total = torch.zeros((), device=device, dtype=torch.float32)
+        # Pre-allocate grad buffers; filled chunk by chunk (recompute-free
+        # backward: we already know dlogits up to the 1/N * upstream scale).
+        dh = torch.empty_like(hidden)
+        dw = torch.zeros_like(weight, dtype=torch.float32)
+        for s in range(0, N, _CHUNK):
+            e = min(s + _CHUNK, N)
+            h = hidden[s:e]
+            t = targets[s:e]
+            logits = h @ weight.t()  # [c, V] in h.dtype (bf16 on GPU)
+            if use_kernel:
+                ext = hip_ext()
+                # Computes per-row loss (fp32) and overwrites `logits` with
+                # (softmax - onehot) in logits.dtype.
+                losses = ext.ce_fwd_bwd_inplace(logits, t)
+                dl = logits
+            else:
+                lf = logits.float()
+                losses = F.cross_entropy(lf, t, reduction="none")
+                dl = torch.softmax(lf, dim=-1)
+                dl[torch.arange(e - s, device=device), t] -= 1.0
+                dl = dl.to(logits.dtype)
+            total = total + losses.sum()
+            dh[s:e] = dl @ weight
+            dw += (dl.t() @ h).float()
+        ctx.save_for_backward(dh, dw)
+        ctx.n_rows = N
+        ctx.w_dtype = weight.dtype
+        return total / N
+
+    @staticmethod
+    def backward(ctx, gout):
+        dh, dw = ctx.saved_tensors
+        scale = gout / ctx.n_rows
+        return dh * scale.to(dh.dtype), (dw * scale).to(ctx.w_dtype), None, None
+
+
+def fused_cross_entropy(
+    hidden: torch.Tensor,
+    weight: torch.Tensor,
+    targets: torch.Tensor,
+    impl: str = "fused",
+) -> torch.Tensor:
+    """Mean CE of logits = hidden @ weight.T against targets.
+
+    hidden: [N, D] (bf16/fp32), weight: [V, D] (tied wte), targets: [N] long.
+    """
+    use_kernel = impl == "fused" and use_hip(hidden)
+    return _FusedLinearCrossEntropy.apply(
+        hidden.contiguous(), weight, targets.contiguous(), use_kernel
+    )
+
+
+def reference_cross_entropy_fp32(hidden, weight, targets):
+    logits = hidden.float() @ weight.float().t()
+    return F.cross_entropy(logits, targets)

@@ -1,0 +1,61 @@
+"""Fused LayerNorm — CDNA4 HIP kernel (one workgroup per row, wave64
+reductions) with a PyTorch fallback.
+
+Replaces the reference's llm-foundry LPLayerNorm / torch LN CUDA kernels
+(SURVEY.md §2.3). Forward computes mean/rstd with fp32 accumulation from
+bf16 inputs; backward is the standard two-reduction LN gradient, fused into
+one kernel per row plus a deterministic column-reduction for dweight/dbias.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import use_hip, hip_ext
+
+
+class _LayerNormHIP(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        ext = hip_ext()
+        y, mean, rstd = ext.layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        ext = hip_ext()
+        dx, dw, db = ext.layernorm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, (db if ctx.has_bias else None), None
+
+
+class FusedLayerNorm(nn.Module):
+    def __init__(self, normalized_shape: int, eps: float = 1e-5, bias: bool = True):
+        super().__init__()
+        self.normalized_shape = (normalized_shape,)
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(normalized_shape))
+        self.bias = nn.Parameter(torch.zeros(normalized_shape)) if bias else None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if use_hip(x):
+            return _LayerNormHIP.apply(
+                x.contiguous(), self.weight, self.bias, self.eps
+            )
+        # Low-precision LN semantics (llm-foundry LPLayerNorm): compute in
+        # fp32, return in input dtype.
+        out = F.layer_norm(
+            x.float(),
+            self.normalized_shape,
+            self.weight.float(),
+            self.bias.float() if self.bias is not None else None,
+            self.eps,
+        )
+        return out.to(x.dtype)
+
+    def extra_repr(self) -> str:
+        return f"{self.normalized_shape}, eps={self.eps}, bias={self.bias is not None}"

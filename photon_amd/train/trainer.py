@@ -1,0 +1,206 @@
+"""Minimal Composer-equivalent trainer, MI355X-native.
+
+Re-implements the slice of MosaicML Composer the reference depends on
+(photon/clients/trainer_utils.py:1117-1721): duration-based fit
+(``local_steps`` in batches), gradient accumulation from
+``global_train_batch_size`` / world size / ``device_train_microbatch_size``,
+bf16 autocast (``precision: amp_bf16``), global-norm gradient clipping,
+cosine-warmup scheduling stepped per optimization batch, timestamps,
+eval loop, and ``ep{e}-ba{b}-rank{r}.pt`` checkpoints with
+``load_ignore_keys`` glob semantics.
+
+No DDP wrapper here: gradient synchronization for the inner data-parallel
+degree is an explicit bucketed RCCL all-reduce in photon_amd.fed.runtime —
+in the 1-GPU-per-client target config it is a no-op.
+"""
+
+from __future__ import annotations
+
+import contextlib
+import fnmatch
+import time
+from pathlib import Path
+
+import torch
+
+from ..conf.schema import duration_to_batches
+from ..ops.clip import clip_grad_norm_
+from ..ops.optim import build_optimizer
+from .scheduler import build_scheduler
+from .timestamp import Timestamp
+
+
+class Trainer:
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        llm_config: dict,
+        train_loader=None,
+        eval_loader=None,
+        device: str | torch.device | None = None,
+        world_size: int = 1,
+        grad_sync_hook=None,
+        run_name: str = "run",
+        save_folder: str | None = None,
+        rank: int = 0,
+    ):
+        self.cfg = llm_config
+        self.device = torch.device(
+            device
+            if device is not None
+            else ("cuda" if torch.cuda.is_available() else "cpu")
+        )
+        self.model = model.to(self.device)
+        self.train_loader = train_loader
+        self.eval_loader = eval_loader
+        self.world_size = world_size
+        self.rank = rank
+        self.grad_sync_hook = grad_sync_hook  # called after backward, before clip
+        self.run_name = run_name
+        self.save_folder = Path(save_folder) if save_folder else None
+
+        self.optimizer = build_optimizer(llm_config["optimizer"], self.model.parameters())
+        self.scheduler = build_scheduler(llm_config, self.optimizer)
+        self.timestamp = Timestamp()
+
+        self.precision = str(llm_config.get("precision", "amp_bf16"))
+        self.microbatch = int(llm_config.get("device_train_microbatch_size", 8))
+        self.global_batch = int(llm_config.get("global_train_batch_size", 256))
+        clip_cfg = (llm_config.get("algorithms") or {}).get("gradient_clipping") or {}
+        self.clip_norm = (
+            float(clip_cfg.get("clipping_threshold", 0.0))
+            if clip_cfg.get("clipping_type", "norm") == "norm"
+            else 0.0
+        )
+        self.max_duration = duration_to_batches(llm_config.get("max_duration", "1000000ba"))
+        self.metrics: dict[str, float] = {}
+
+    # -- precision ----------------------------------------------------------
+    def autocast(self):
+        if self.precision == "amp_bf16":
+            return torch.autocast(device_type=self.device.type, dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
+    @property
+    def grad_accum(self) -> int:
+        per_device = max(self.global_batch // self.world_size, self.microbatch)
+        return max(per_device // self.microbatch, 1)
+
+    # -- train --------------------------------------------------------------
+    def train_batch(self, batches: list[dict]) -> float:
+        """One optimization batch = grad_accum microbatches. Returns loss."""
+        self.optimizer.zero_grad(set_to_none=True)
+        total_loss = 0.0
+        n = len(batches)
+        for mb in batches:
+            ids = mb["input_ids"].to(self.device, non_blocking=True)
+            with self.autocast():
+                out = self.model(ids, labels=ids)
+            loss = out["loss"] / n
+            loss.backward()
+            total_loss += float(loss.detach())
+        if self.grad_sync_hook is not None:
+            self.grad_sync_hook(self.model)
+        if self.clip_norm > 0:
+            clip_grad_norm_(list(self.model.parameters()), self.clip_norm)
+        self.scheduler.step(self.timestamp.batch + 1)
+        self.optimizer.step()
+        samples = sum(b["input_ids"].shape[0] for b in batches) * self.world_size
+        tokens = samples * batches[0]["input_ids"].shape[1]
+        self.timestamp.tick_batch(samples, tokens)
+        return total_loss
+
+    def fit(self, duration_batches: int | str, callback=None) -> dict:
+        """Train for `duration_batches` optimization batches; returns metrics."""
+        n_batches = duration_to_batches(duration_batches)
+        t0 = time.time()
+        losses = []
+        for _ in range(n_batches):
+            if self.timestamp.batch >= self.max_duration:
+                break
+            mbs = [
+                self.train_loader.next_batch() for _ in range(self.grad_accum)
+            ]
+            loss = self.train_batch(mbs)
+            losses.append(loss)
+            if callback is not None:
+                callback(self, loss)
+        fit_time = time.time() - t0
+        self.metrics.update(
+            {
+                "loss/train/total": losses[-1] if losses else float("nan"),
+                "time/train": fit_time,
+                "throughput/batches_per_sec": (len(losses) / fit_time) if fit_time > 0 else 0.0,
+            }
+        )
+        return dict(self.metrics)
+
+    # -- eval ---------------------------------------------------------------
+    @torch.no_grad()
+    def eval(self, subset_num_batches: int = -1) -> dict:
+        if self.eval_loader is None:
+            return {}
+        self.model.eval()
+        n = subset_num_batches if subset_num_batches > 0 else 8
+        total, count = 0.0, 0
+        for _ in range(n):
+            batch = self.eval_loader.next_batch()
+            ids = batch["input_ids"].to(self.device, non_blocking=True)
+            with self.autocast():
+                out = self.model(ids, labels=ids)
+            total += float(out["loss"])
+            count += 1
+        self.model.train()
+        loss = total / max(count, 1)
+        metrics = {
+            "metrics/eval/LanguageCrossEntropy": loss,
+            "metrics/eval/LanguagePerplexity": float(torch.exp(torch.tensor(loss))),
+            "eval_samples": count * self.eval_loader.batch_size,
+        }
+        self.metrics.update(metrics)
+        return metrics
+
+    # -- checkpoint (ep{e}-ba{b}-rank{r}.pt, Composer-compatible keys) ------
+    def checkpoint_name(self) -> str:
+        return f"ep{self.timestamp.epoch}-ba{self.timestamp.batch}-rank{self.rank}.pt"
+
+    def save_checkpoint(self, folder: str | Path | None = None) -> Path:
+        folder = Path(folder or self.save_folder or ".")
+        folder.mkdir(parents=True, exist_ok=True)
+        path = folder / self.checkpoint_name()
+        state = {
+            "state": {
+                "model": self.model.state_dict(),
+                "optimizers": {type(self.optimizer).__name__: self.optimizer.state_dict()},
+                "schedulers": {"lr": self.scheduler.state_dict()},
+                "timestamp": self.timestamp.state_dict(),
+                "dataset_state": (
+                    self.train_loader.state_dict() if self.train_loader else {}
+                ),
+                "run_name": self.run_name,
+            },
+            "rng": {"torch": torch.get_rng_state()},
+        }
+        torch.save(state, path)
+        return path
+
+    def load_checkpoint(
+        self, path: str | Path, load_ignore_keys: list[str] | None = None
+    ) -> None:
+        ckpt = torch.load(path, map_location=self.device, weights_only=False)
+        state = ckpt["state"]
+        ignore = load_ignore_keys or []
+
+        def ignored(key: str) -> bool:
+            return any(fnmatch.fnmatch(key, pat) for pat in ignore)
+
+        self.model.load_state_dict(state["model"])
+        if not ignored("*optim*") and "optimizers" in state:
+            (opt_state,) = state["optimizers"].values()
+            self.optimizer.load_state_dict(opt_state)
+        if not ignored("*scheduler*") and "schedulers" in state:
+            self.scheduler.load_state_dict(state["schedulers"]["lr"])
+        if not ignored("*dataset_state*") and self.train_loader is not None:
+            self.train_loader.load_state_dict(state.get("dataset_state", {}))
+        if not ignored("*timestamp*"):
+            self.timestamp.load_state_dict(state["timestamp"])
