@@ -135,9 +135,21 @@ def _load_yaml(path: Path, *, allow_list: bool = False) -> Any:
 
 def _parse_value(text: str) -> Any:
     try:
-        return yaml.safe_load(text)
+        value = yaml.safe_load(text)
     except yaml.YAMLError:
         return text
+    if isinstance(value, str):
+        # YAML 1.1 misses bare scientific notation ("1e-5"); Hydra treats it
+        # as a float — match that.
+        try:
+            return int(value)
+        except ValueError:
+            pass
+        try:
+            return float(value)
+        except ValueError:
+            pass
+    return value
 
 
 _INTERP_RE = re.compile(r"\$\{([A-Za-z0-9_.@-]+)\}")

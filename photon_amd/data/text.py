@@ -21,7 +21,8 @@ def _stream_local(stream_group_entry: dict) -> str | None:
     return stream_cfg.get("local")
 
 
-def _build_dataset(ds_cfg: dict, split_cfg: dict, seq_len: int, client_id, split, seed):
+def _build_dataset(ds_cfg: dict, split_cfg: dict, seq_len: int, client_id, split, seed,
+                   vocab_size: int = 50368):
     streams = split_cfg.get("streams")
     synthetic = bool(split_cfg.get("synthetic", False))
     local = None
@@ -43,6 +44,7 @@ def _build_dataset(ds_cfg: dict, split_cfg: dict, seq_len: int, client_id, split
             )
     return SyntheticTokenDataset(
         seq_len,
+        vocab_size=vocab_size,
         seed=seed,
         client_id=client_id if client_id is not None else 0,
         split=str(split_cfg.get("split", split)),
@@ -55,7 +57,8 @@ def build_train_loader(cfg, client_id=None, batch_size=None) -> StatefulLoader:
     split_cfg = cfg["dataset"]["train"]
     bs = batch_size or int(llm.get("device_train_microbatch_size", 8))
     ds = _build_dataset(cfg["dataset"], split_cfg, seq_len, client_id, "train",
-                        int(cfg.get("seed", 1337)))
+                        int(cfg.get("seed", 1337)),
+                        vocab_size=int(llm["model"].get("vocab_size", 50368)))
     return StatefulLoader(ds, bs)
 
 
@@ -65,5 +68,6 @@ def build_eval_loader(cfg, client_id=None, batch_size=None) -> StatefulLoader:
     split_cfg = cfg["dataset"]["val"]
     bs = batch_size or int(llm.get("device_eval_batch_size", 8))
     ds = _build_dataset(cfg["dataset"], split_cfg, seq_len, client_id, "validation",
-                        int(cfg.get("seed", 1337)))
+                        int(cfg.get("seed", 1337)),
+                        vocab_size=int(llm["model"].get("vocab_size", 50368)))
     return StatefulLoader(ds, bs)

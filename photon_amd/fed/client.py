@@ -1,0 +1,149 @@
+"""Client-side local training/eval — the reference's llm_fit / llm_eval
+(photon/clients/llm_client_functions.py:53-353) re-shaped for one process
+per GPU (no NodeManager/Worker/SHM mailboxes — SURVEY.md §3.2 collapses to
+a function call)."""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+
+import torch
+
+from ..conf.schema import duration_to_batches
+from ..data import build_eval_loader, build_train_loader
+from ..models import build_model
+from ..train import Trainer
+from .flat import FlatParams
+
+
+@dataclass
+class ClientState:
+    """Reference photon/utils.py:41-53."""
+
+    cid: int
+    steps_done: int = 0
+    metrics: dict = field(default_factory=dict)
+
+
+class FedClient:
+    """A persistent per-rank client: owns the model + trainer across rounds
+    (the reference's persistent-Trainer reuse path, trainer_utils.py:330-653),
+    re-seeded per cid for multi-client-per-rank rounds."""
+
+    def __init__(self, cfg, device, rank: int = 0, world_size_inner: int = 1):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        llm = cfg["llm_config"]
+        torch.manual_seed(int(llm.get("seed", 17)))
+        self.model = build_model(llm)
+        self.trainer: Trainer | None = None
+        self.rank = rank
+        self.client_states: dict[int, ClientState] = {}
+        # Per-client persistent trainer state (the reference restores the
+        # client's timestamp + dataset_state at every fit,
+        # llm_client_functions.py:163-175) so results are identical whether a
+        # client runs on its own rank or shares one.
+        self._timestamps: dict[int, dict] = {}
+        self._loader_states: dict[int, dict] = {}
+
+    def _ensure_trainer(self, cid: int) -> Trainer:
+        llm = self.cfg["llm_config"]
+        if self.trainer is None:
+            self.trainer = Trainer(
+                self.model,
+                llm,
+                train_loader=build_train_loader(self.cfg, client_id=cid),
+                eval_loader=build_eval_loader(self.cfg, client_id=cid),
+                device=self.device,
+                run_name=str(self.cfg.get("run_uuid", "run")),
+                rank=self.rank,
+            )
+        else:
+            # switch streams when this rank picks up a different client id
+            self.trainer.train_loader = build_train_loader(self.cfg, client_id=cid)
+            self.trainer.eval_loader = build_eval_loader(self.cfg, client_id=cid)
+        return self.trainer
+
+    def fit(
+        self,
+        cid: int,
+        global_flat: torch.Tensor,
+        layout: FlatParams,
+        server_round: int,
+        local_steps=None,
+        reset_optimizer: bool = True,
+    ) -> tuple[torch.Tensor, float, dict]:
+        """Run local_steps batches from the global params.
+
+        Returns (local flat params fp32, n_samples, metrics). n_samples =
+        local_steps * global_train_batch_size (reference
+        post_process_client_result, clients/utils.py:514-652).
+        """
+        llm = self.cfg["llm_config"]
+        steps = duration_to_batches(
+            local_steps if local_steps is not None else llm.get("local_steps", "500ba")
+        )
+        trainer = self._ensure_trainer(cid)
+        # restore per-client timestamp + dataset state
+        from ..train.timestamp import Timestamp
+
+        trainer.timestamp = Timestamp()
+        if cid in self._timestamps:
+            trainer.timestamp.load_state_dict(self._timestamps[cid])
+        if cid in self._loader_states and not self.cfg["fl"].get("reset_dataset_state", False):
+            trainer.train_loader.load_state_dict(self._loader_states[cid])
+        t0 = time.time()
+        # set params from the global buffer (HBM->HBM copies, no host hop)
+        views = layout.layer_views_of(global_flat)
+        params = dict(self.model.named_parameters())
+        with torch.no_grad():
+            for n, v in zip(layout.names, views):
+                params[n].data.copy_(v.to(params[n].dtype))
+        set_params_time = time.time() - t0
+
+        if reset_optimizer:
+            trainer.optimizer.state.clear()
+
+        t1 = time.time()
+        fit_metrics = trainer.fit(steps)
+        fit_time = time.time() - t1
+
+        t2 = time.time()
+        local_flat = torch.zeros_like(global_flat)
+        out_views = layout.layer_views_of(local_flat)
+        params = dict(self.model.named_parameters())
+        with torch.no_grad():
+            for n, v in zip(layout.names, out_views):
+                v.copy_(params[n].detach().to(torch.float32))
+        get_params_time = time.time() - t2
+
+        n_samples = float(steps * int(llm.get("global_train_batch_size", 256)))
+        st = self.client_states.setdefault(cid, ClientState(cid))
+        st.steps_done += steps
+        self._timestamps[cid] = trainer.timestamp.state_dict()
+        self._loader_states[cid] = trainer.train_loader.state_dict()
+        metrics = {
+            "client/fit_set_parameters_time": set_params_time,
+            "client/fit_time": fit_time,
+            "client/fit_get_parameters_time": get_params_time,
+            "loss/train/total": fit_metrics.get("loss/train/total", float("nan")),
+            "steps_done": st.steps_done,
+        }
+        return local_flat, n_samples, metrics
+
+    @torch.no_grad()
+    def evaluate(
+        self, cid: int, global_flat: torch.Tensor, layout: FlatParams,
+        subset_num_batches: int = 8,
+    ) -> tuple[float, float, dict]:
+        """Returns (eval_loss, n_samples, metrics) — reference llm_eval."""
+        trainer = self._ensure_trainer(cid)
+        views = layout.layer_views_of(global_flat)
+        params = dict(self.model.named_parameters())
+        for n, v in zip(layout.names, views):
+            params[n].data.copy_(v.to(params[n].dtype))
+        metrics = trainer.eval(subset_num_batches)
+        loss = metrics.get("metrics/eval/LanguageCrossEntropy", float("nan"))
+        n = float(metrics.get("eval_samples", 0))
+        return loss, n, metrics

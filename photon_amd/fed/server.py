@@ -1,0 +1,261 @@
+"""The federated round loop — reference server_app.main (photon/server_app.py:85-424)
+re-designed for the RCCL-symmetric topology.
+
+Every rank executes the same loop; there is no separate server process.
+Rank 0 additionally owns checkpointing and history. Per round:
+
+  1. sample clients       — replicated seeded RNG (no control message),
+  2. local fit            — each rank trains its assigned client ids for
+                            local_steps (the NodeManager work-queue collapses
+                            to a per-rank loop),
+  3. aggregation          — ONE RCCL all-reduce of the n_i/sum(n)-scaled flat
+                            buffers (photon/server/fit_utils.py:41-217 +
+                            strategy/aggregation.py in a single collective),
+  4. server-opt update    — the FedOpt strategy applied redundantly on every
+                            rank (deterministic => replicas stay bit-identical;
+                            replaces the post-update re-broadcast at
+                            server_app.py:327 with zero communication),
+  5. eval every eval_period — weighted_loss_avg over ranks,
+  6. rank-0 server checkpoint + retention cleanup.
+
+Failure semantics parity (fit_utils.py:198-288): a client whose local fit
+raises contributes zero weight; failure counts are all-reduced so every rank
+agrees; more than accept_failures_cnt failures raises TooManyFailuresError
+unless ignore_failed_rounds.
+"""
+
+from __future__ import annotations
+
+import time
+from pathlib import Path
+
+import torch
+
+from ..conf.schema import duration_to_batches
+from ..history import History
+from ..models import build_model
+from .client import FedClient
+from .flat import FlatParams
+from .noise_scale import FedSimpleNoiseScale
+from .runtime import Comm, assign_clients_to_ranks, sample_clients
+from .server_ckpt import (
+    interpret_resume_round,
+    obtain_sorted_rounds,
+    resume_from_round,
+    upload_server_checkpoint,
+)
+from .strategies import dispatch_strategy
+
+
+class TooManyFailuresError(RuntimeError):
+    pass
+
+
+def weighted_loss_avg(losses_and_weights: list[tuple[float, float]]) -> float:
+    total = sum(w for _, w in losses_and_weights)
+    if total == 0:
+        return float("nan")
+    return sum(l * w for l, w in losses_and_weights) / total
+
+
+class FedServer:
+    """Holds the symmetric round-loop state for one rank."""
+
+    def __init__(self, cfg, comm: Comm, device):
+        self.cfg = cfg
+        self.comm = comm
+        self.device = torch.device(device)
+        fl = cfg["fl"]
+        self.n_total = int(fl["n_total_clients"])
+        self.n_per_round = int(fl["n_clients_per_round"])
+        self.n_rounds = int(fl["n_rounds"])
+        self.eval_period = int(fl.get("eval_period", 1))
+        self.accept_failures_cnt = int(fl.get("accept_failures_cnt", 0))
+        self.ignore_failed_rounds = bool(fl.get("ignore_failed_rounds", False))
+        self.seed = int(cfg.get("seed", 1337))
+        self.run_uuid = str(cfg.get("run_uuid", "run"))
+        self.saving_path = Path(
+            cfg["photon"].get("saving_path") or "checkpoints"
+        )
+        self.checkpoint_enabled = bool(cfg["photon"].get("checkpoint", False))
+
+        self.client = FedClient(cfg, self.device, rank=comm.rank)
+        self.layout = FlatParams(
+            self.client.model,
+            filter_key=(
+                str(fl.get("set_trainer_key_to_filter", "transformer"))
+                if fl.get("set_trainer_params_filter_keys", True)
+                else None
+            ),
+            device=self.device,
+        )
+        self.strategy = dispatch_strategy(
+            fl.get("strategy_name", "NESTOROV"), self.layout, fl.get("strategy_kwargs")
+        )
+        self.noise_scale = (
+            FedSimpleNoiseScale(beta=float(fl.get("noise_scale_beta", 0.99)))
+            if fl.get("use_noise_scale_metric", False)
+            else None
+        )
+        self.history = History(
+            run_dir=self.saving_path / self.run_uuid if comm.rank == 0 else None,
+            use_wandb=bool(cfg.get("use_wandb", False)),
+            wandb_setup=dict(cfg.get("wandb", {}).get("setup", {})) if cfg.get("use_wandb") else None,
+            suffix="_server",
+        )
+        self.server_steps_cumulative = 0
+        self.start_round = 1
+
+    # -- initialization / resume -------------------------------------------
+    def initialize(self) -> None:
+        """initialize_round / resume_from_round (init_utils.py:128-287)."""
+        resumed = None
+        resume_round = self.cfg["photon"].get("resume_round", -1)
+        rounds = obtain_sorted_rounds(
+            self.saving_path, self.run_uuid, self.strategy.state_keys
+        )
+        target = interpret_resume_round(resume_round, rounds)
+        if target is not None:
+            state = resume_from_round(
+                self.saving_path, self.run_uuid, target, self.strategy, self.layout
+            )
+            self.server_steps_cumulative = int(state.get("server_steps_cumulative", 0))
+            self.history.load_state(state.get("history", {}))
+            self.start_round = target + 1
+            resumed = target
+        else:
+            # fresh init: rank 0's model init is the global model; broadcast
+            # the flat buffer once (the only full-parameter broadcast).
+            self.layout.copy_from_model(self.client.model)
+            self.comm.broadcast_flat(self.layout.flat, src=0)
+            self.strategy.initialize(self.layout.flat)
+        # all ranks start from identical global params
+        self.comm.broadcast_flat(self.strategy.params, src=0)
+        if resumed is not None and self.comm.rank == 0:
+            print(f"[fed] resumed from round {resumed}")
+
+    # -- one round -----------------------------------------------------------
+    def run_round(self, server_round: int) -> dict:
+        t_round = time.time()
+        fl = self.cfg["fl"]
+        sampled = sample_clients(self.seed, server_round, self.n_total, self.n_per_round)
+        assignment = assign_clients_to_ranks(sampled, self.comm.world_size)
+        my_cids = assignment[self.comm.rank]
+
+        local_sum = torch.zeros_like(self.strategy.params)
+        local_weight = 0.0
+        failures = 0
+        steps_done_max = 0
+        fit_metrics: dict = {}
+        per_client_sq_norms: list[tuple[float, float]] = []  # (n_i, ||g_i||^2)
+
+        t_fit = time.time()
+        for cid in my_cids:
+            try:
+                local_flat, n_samples, metrics = self.client.fit(
+                    cid,
+                    self.strategy.params,
+                    self.layout,
+                    server_round,
+                    reset_optimizer=bool(fl.get("reset_optimizer", True)),
+                )
+                local_sum.add_(local_flat, alpha=n_samples)
+                local_weight += n_samples
+                steps_done_max = max(steps_done_max, int(metrics.get("steps_done", 0)))
+                fit_metrics = metrics
+                if self.noise_scale is not None:
+                    g = self.strategy.params - local_flat
+                    per_client_sq_norms.append(
+                        (n_samples, float(torch.dot(g, g)))
+                    )
+            except Exception as e:  # failure budget semantics
+                failures += 1
+                if self.comm.rank == 0 or True:
+                    print(f"[fed] client {cid} fit failed: {e!r}")
+        fit_time = time.time() - t_fit
+
+        # agree on failures across ranks
+        fail_total = sum(self.comm.all_gather_scalars(float(failures)))
+        if fail_total > self.accept_failures_cnt and not self.ignore_failed_rounds:
+            raise TooManyFailuresError(
+                f"round {server_round}: {int(fail_total)} client failures "
+                f"(accept_failures_cnt={self.accept_failures_cnt})"
+            )
+
+        # ONE weighted all-reduce over xGMI
+        t_agg = time.time()
+        fedavg_flat, total_weight = self.comm.weighted_average_(local_sum, local_weight)
+        agg_time = time.time() - t_agg
+        if total_weight == 0:
+            raise TooManyFailuresError(f"round {server_round}: no successful clients")
+
+        # replicated server-opt update
+        strat_metrics = self.strategy.update(
+            fedavg_flat, server_round, len(sampled)
+        )
+        if self.noise_scale is not None:
+            g_big_sq = float(strat_metrics.get("l2_norm_pseudo_gradient", 0.0)) ** 2
+            ns = self.noise_scale.update(per_client_sq_norms, g_big_sq, self.comm)
+            strat_metrics.update(ns)
+
+        # steps bookkeeping: server_steps_cumulative += max(steps_done)
+        steps_max_global = max(
+            self.comm.all_gather_scalars(float(steps_done_max))
+        )
+        self.server_steps_cumulative = int(steps_max_global)
+
+        round_metrics = {
+            "server/fit_round_time": fit_time,
+            "server/aggregate_time": agg_time,
+            "server/round_time": time.time() - t_round,
+            "server/sampled_clients": len(sampled),
+            "server/failures": fail_total,
+            "server_steps_cumulative": self.server_steps_cumulative,
+            **{k: v for k, v in strat_metrics.items() if not isinstance(v, list)},
+            **{k: v for k, v in fit_metrics.items() if isinstance(v, (int, float))},
+        }
+        if self.comm.rank == 0:
+            self.history.add_metrics_distributed(server_round, round_metrics)
+        return round_metrics
+
+    def evaluate_round(self, server_round: int) -> float:
+        """Reference evaluate_round (evaluate_utils.py:232): every rank
+        evaluates its client shard; weighted_loss_avg across ranks."""
+        subset = int(self.cfg["llm_config"].get("eval_subset_num_batches", -1))
+        if subset <= 0:
+            subset = 8
+        cid = self.comm.rank if self.comm.rank < self.n_total else 0
+        loss, n, metrics = self.client.evaluate(
+            cid, self.strategy.params, self.layout, subset
+        )
+        losses = self.comm.all_gather_scalars(loss)
+        weights = self.comm.all_gather_scalars(n)
+        avg = weighted_loss_avg(list(zip(losses, weights)))
+        if self.comm.rank == 0:
+            self.history.add_loss_distributed(server_round, avg)
+            self.history.add_metrics_distributed(
+                server_round, {"metrics/eval/LanguageCrossEntropy_avg": avg}
+            )
+        return avg
+
+    # -- full loop -----------------------------------------------------------
+    def run(self, n_rounds: int | None = None) -> History:
+        self.initialize()
+        last = self.start_round + (n_rounds or self.n_rounds) - 1
+        for r in range(self.start_round, last + 1):
+            self.run_round(r)
+            if self.eval_period > 0 and r % self.eval_period == 0:
+                self.evaluate_round(r)
+            if self.checkpoint_enabled and self.comm.rank == 0:
+                upload_server_checkpoint(
+                    self.saving_path,
+                    self.run_uuid,
+                    r,
+                    self.strategy,
+                    self.layout,
+                    self.history.state(),
+                    {cid: vars(st) for cid, st in self.client.client_states.items()},
+                    self.server_steps_cumulative,
+                )
+            self.comm.barrier()
+        return self.history

@@ -1,0 +1,150 @@
+"""Server round checkpoints — bit-compatible with the reference layout.
+
+Layout per round (photon/server/s3_utils.py:348-548; key names
+photon/strategy/constants.py:4-6):
+
+    {saving_path}/{run_uuid}/server/{round}/state.bin
+        pickle: {server_round, history, time_offset, client_state (as str),
+                 server_steps_cumulative}
+    {saving_path}/{run_uuid}/server/{round}/current_server_parameters.npz
+    {saving_path}/{run_uuid}/server/{round}/current_momentum_vector.npz
+    {saving_path}/{run_uuid}/server/{round}/current_second_momentum_vector.npz
+
+Resume: ``photon.resume_round`` (negative = index from the latest), rounds
+enumerated by requiring all of the strategy's state_keys present
+(s3_utils.py:215-272, 1261-1318). The store is the local FS here; the
+``list_objects`` local-or-S3 abstraction survives in fed/store.py.
+"""
+
+from __future__ import annotations
+
+import pickle
+from pathlib import Path
+
+import numpy as np
+import torch
+
+from .flat import FlatParams
+from .strategies import STATE_M1, STATE_M2, STATE_PARAMS, Strategy
+
+_KEY_FILES = {
+    STATE_PARAMS: "current_server_parameters.npz",
+    STATE_M1: "current_momentum_vector.npz",
+    STATE_M2: "current_second_momentum_vector.npz",
+}
+
+
+def server_dir(saving_path: str | Path, run_uuid: str) -> Path:
+    return Path(saving_path) / str(run_uuid) / "server"
+
+
+def upload_server_checkpoint(
+    saving_path,
+    run_uuid: str,
+    server_round: int,
+    strategy: Strategy,
+    layout: FlatParams,
+    history: dict,
+    client_state: dict,
+    server_steps_cumulative: int,
+    time_offset: float = 0.0,
+) -> Path:
+    rd = server_dir(saving_path, run_uuid) / str(server_round)
+    rd.mkdir(parents=True, exist_ok=True)
+    state = {
+        "server_round": int(server_round),
+        "history": history,
+        "time_offset": float(time_offset),
+        "client_state": str(client_state),
+        "server_steps_cumulative": int(server_steps_cumulative),
+    }
+    with open(rd / "state.bin", "wb") as f:
+        pickle.dump(state, f)
+    for key, tensor in strategy.state_tensors().items():
+        layout.save_npz(rd / _KEY_FILES[key], tensor)
+    return rd
+
+
+def obtain_sorted_rounds(saving_path, run_uuid: str, state_keys) -> list[int]:
+    """Rounds with a complete checkpoint (all state_keys + state.bin),
+    ascending (s3_utils.py:1261-1318)."""
+    base = server_dir(saving_path, run_uuid)
+    if not base.exists():
+        return []
+    rounds = []
+    for d in base.iterdir():
+        if not d.name.isdigit():
+            continue
+        need = [d / "state.bin"] + [d / _KEY_FILES[k] for k in state_keys]
+        if all(p.exists() for p in need):
+            rounds.append(int(d.name))
+    return sorted(rounds)
+
+
+def interpret_resume_round(resume_round: int | None, rounds: list[int]) -> int | None:
+    """Reference semantics (s3_utils.py:215-272): positive = that round,
+    negative = index from the latest (-1 = latest), None/0 or no rounds = no resume."""
+    if not rounds or resume_round in (None, 0):
+        return None
+    if resume_round > 0:
+        return resume_round if resume_round in rounds else None
+    try:
+        return rounds[resume_round]
+    except IndexError:
+        return None
+
+
+def resume_from_round(
+    saving_path, run_uuid: str, server_round: int, strategy: Strategy, layout: FlatParams
+) -> dict:
+    """Load strategy state + server state from a round dir; returns state.bin."""
+    rd = server_dir(saving_path, run_uuid) / str(server_round)
+    with open(rd / "state.bin", "rb") as f:
+        state = pickle.load(f)
+    tensors = {}
+    for key in strategy.state_keys:
+        arrays = layout.load_npz(rd / _KEY_FILES[key])
+        flat = torch.cat(
+            [torch.from_numpy(np.ascontiguousarray(a, dtype=np.float32)).reshape(-1) for a in arrays]
+        ).to(strategy.params.device)
+        tensors[key] = flat
+    strategy.load_state_tensors(tensors)
+    return state
+
+
+def delete_rounds(saving_path, run_uuid: str, keep_rounds: list[int]) -> None:
+    """Retention cleanup (s3_utils.py:1321-1641): remove round dirs not in keep."""
+    import shutil
+
+    base = server_dir(saving_path, run_uuid)
+    if not base.exists():
+        return
+    for d in base.iterdir():
+        if d.name.isdigit() and int(d.name) not in keep_rounds:
+            shutil.rmtree(d)
+
+
+def copy_old_checkpoints_to_new_run(
+    saving_path, old_run_uuid: str, new_run_uuid: str, state_keys,
+    copy_client_checkpoints: bool = True,
+) -> int | None:
+    """Cross-run restore (s3_utils.py:275-345,1478-1608): copy the latest
+    complete server round (and client checkpoint dirs) into the new run.
+    Returns the restored round or None."""
+    import shutil
+
+    rounds = obtain_sorted_rounds(saving_path, old_run_uuid, state_keys)
+    if not rounds:
+        return None
+    latest = rounds[-1]
+    src = server_dir(saving_path, old_run_uuid) / str(latest)
+    dst = server_dir(saving_path, new_run_uuid) / str(latest)
+    dst.mkdir(parents=True, exist_ok=True)
+    for f in src.iterdir():
+        shutil.copy2(f, dst / f.name)
+    if copy_client_checkpoints:
+        old_base = Path(saving_path) / str(old_run_uuid)
+        new_base = Path(saving_path) / str(new_run_uuid)
+        for cdir in old_base.glob("client_*"):
+            shutil.copytree(cdir, new_base / cdir.name, dirs_exist_ok=True)
+    return latest

@@ -1,0 +1,65 @@
+"""Data layer tests: shard roundtrip, resume state, synthetic determinism."""
+
+import numpy as np
+import torch
+
+from photon_amd.data.shards import StatefulLoader, TokenShardDataset, TokenShardWriter
+from photon_amd.data.synthetic import SyntheticTokenDataset
+
+
+def test_shard_writer_reader_roundtrip(tmp_path):
+    w = TokenShardWriter(tmp_path / "d", tokens_per_shard=100)
+    all_tokens = np.arange(350, dtype=np.uint32)
+    w.write(all_tokens[:120])
+    w.write(all_tokens[120:])
+    w.close()
+    ds = TokenShardDataset(tmp_path / "d", seq_len=16)
+    assert len(ds) == 350 // 16
+    # windows must reproduce the stream exactly, incl. shard-straddling ones
+    for i in range(len(ds)):
+        got = ds[i].numpy()
+        np.testing.assert_array_equal(got, np.arange(i * 16, i * 16 + 16))
+
+
+def test_shard_shuffle_deterministic(tmp_path):
+    w = TokenShardWriter(tmp_path / "d", tokens_per_shard=64)
+    w.write(np.arange(256, dtype=np.uint32))
+    w.close()
+    a = TokenShardDataset(tmp_path / "d", 8, shuffle=True, shuffle_seed=1)
+    b = TokenShardDataset(tmp_path / "d", 8, shuffle=True, shuffle_seed=1)
+    c = TokenShardDataset(tmp_path / "d", 8, shuffle=True, shuffle_seed=2)
+    assert torch.equal(a[0], b[0])
+    assert (a.order != c.order).any()
+
+
+def test_stateful_loader_resume(tmp_path):
+    ds = SyntheticTokenDataset(8, vocab_size=64, seed=3)
+    l1 = StatefulLoader(ds, 4)
+    b1 = l1.next_batch()
+    state = l1.state_dict()
+    b2 = l1.next_batch()
+
+    l2 = StatefulLoader(ds, 4)
+    l2.load_state_dict(state)
+    b2b = l2.next_batch()
+    assert torch.equal(b2["input_ids"], b2b["input_ids"])
+    assert not torch.equal(b1["input_ids"], b2["input_ids"])
+
+
+def test_synthetic_determinism_and_isolation():
+    a = SyntheticTokenDataset(16, vocab_size=100, seed=1, client_id=0)
+    b = SyntheticTokenDataset(16, vocab_size=100, seed=1, client_id=0)
+    c = SyntheticTokenDataset(16, vocab_size=100, seed=1, client_id=1)
+    v = SyntheticTokenDataset(16, vocab_size=100, seed=1, client_id=0, split="validation")
+    assert torch.equal(a[5], b[5])
+    assert not torch.equal(a[5], c[5])
+    assert not torch.equal(a[5], v[5])
+    assert int(a[5].max()) < 100
+
+
+def test_synthetic_zipf_shape():
+    ds = SyntheticTokenDataset(2048, vocab_size=50368, seed=7)
+    toks = torch.cat([ds[i] for i in range(4)])
+    # Zipf-ish: low ids dominate
+    assert (toks < 1000).float().mean() > 0.3
+    assert int(toks.max()) < 50368
