@@ -1,0 +1,52 @@
+// Torch-extension bindings for the photon_amd CDNA4 HIP kernels.
+// Built in-tree as photon_amd/ops/_photon_hip.so (PYTORCH_ROCM_ARCH=gfx950);
+// the .so travels with the repo snapshot to GPU boxes.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+namespace photon_hip {
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         c10::optional<torch::Tensor> b,
+                                         double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+torch::Tensor ce_fwd_bwd_inplace(torch::Tensor logits, torch::Tensor targets);
+void adamw_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                double lr, double b1, double b2, double eps, double wd,
+                double bc1, double bc2);
+void adopt_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
+                std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
+                double lr, double b1, double b2, double eps, double wd,
+                double clip, long step);
+torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> gs);
+void multi_tensor_scale_clip(std::vector<torch::Tensor> gs,
+                             torch::Tensor total_norm, double max_norm);
+std::vector<torch::Tensor> attn_fwd_launch(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v,
+                                           torch::Tensor slopes, bool causal);
+std::vector<torch::Tensor> attn_bwd_launch(torch::Tensor dout, torch::Tensor q,
+                                           torch::Tensor k, torch::Tensor v,
+                                           torch::Tensor slopes,
+                                           torch::Tensor o, torch::Tensor lse,
+                                           bool causal);
+
+}  // namespace photon_hip
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  using namespace photon_hip;
+  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm forward");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm backward");
+  m.def("ce_fwd_bwd_inplace", &ce_fwd_bwd_inplace,
+        "fused CE loss + in-place softmax gradient");
+  m.def("adamw_step", &adamw_step, "fused multi-tensor AdamW step");
+  m.def("adopt_step", &adopt_step, "fused multi-tensor ADOPT step");
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm);
+  m.def("multi_tensor_scale_clip", &multi_tensor_scale_clip);
+  m.def("attn_fwd", &attn_fwd_launch, "flash attention forward (ALiBi fused)");
+  m.def("attn_bwd", &attn_bwd_launch, "flash attention backward");
+}
