@@ -168,8 +168,8 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_fwd_kernel(
 
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * KB;
-    stage_tile<D>(k, kv0, S, D, k_img + 0, nullptr);
-    stage_tile<D>(v, kv0, S, D, nullptr, vt_img);
+    stage_tile<D>(k + base, kv0, S, D, k_img + 0, nullptr);
+    stage_tile<D>(v + base, kv0, S, D, nullptr, vt_img);
     __syncthreads();
 
     const bool active = !causal || (kv0 <= my_q_max);
@@ -310,8 +310,8 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dq_kernel(
 
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * KB;
-    stage_tile<D>(k, kv0, S, D, k_img, kt_img);
-    stage_tile<D>(v, kv0, S, D, v_img, nullptr);
+    stage_tile<D>(k + base, kv0, S, D, k_img, kt_img);
+    stage_tile<D>(v + base, kv0, S, D, v_img, nullptr);
     __syncthreads();
 
     const bool active = !causal || (kv0 <= my_q_max);
@@ -432,8 +432,8 @@ __global__ __launch_bounds__(ATT_BLOCK) void attn_bwd_dkdv_kernel(
 
   for (int t = t0; t < n_tiles; ++t) {
     const int qt0 = t * QB;
-    stage_tile<D>(q, qt0, S, D, q_img, qt_img);
-    stage_tile<D>(dout, qt0, S, D, do_img, dot_img);
+    stage_tile<D>(q + base, qt0, S, D, q_img, qt_img);
+    stage_tile<D>(dout + base, qt0, S, D, do_img, dot_img);
     for (int i = threadIdx.x; i < 32; i += ATT_BLOCK) {
       const int qi = qt0 + i;
       lse_t[i] = (qi < S) ? lse[bh * (long)S + qi] : INFINITY;

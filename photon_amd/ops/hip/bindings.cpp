@@ -29,6 +29,8 @@ void multi_tensor_scale_clip(std::vector<torch::Tensor> gs,
 std::vector<torch::Tensor> attn_fwd_launch(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v,
                                            torch::Tensor slopes, bool causal);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor BT);
+torch::Tensor pack_probe(torch::Tensor dummy);
 std::vector<torch::Tensor> attn_bwd_launch(torch::Tensor dout, torch::Tensor q,
                                            torch::Tensor k, torch::Tensor v,
                                            torch::Tensor slopes,
@@ -49,4 +51,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale_clip", &multi_tensor_scale_clip);
   m.def("attn_fwd", &attn_fwd_launch, "flash attention forward (ALiBi fused)");
   m.def("attn_bwd", &attn_bwd_launch, "flash attention backward");
+  m.def("mfma_probe", &mfma_probe);
+  m.def("pack_probe", &pack_probe);
 }

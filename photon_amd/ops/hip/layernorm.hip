@@ -17,6 +17,9 @@ namespace photon_hip {
 // ---------------------------------------------------------------------------
 // Forward: y = (x - mean) * rstd * w + b ; saves mean, rstd (fp32 per row)
 // ---------------------------------------------------------------------------
+// D up to 8192 (MPT-7B d_model 4096): x stays in registers (<= 4 chunks of
+// 8 per thread), so mean and variance are exact two-pass fp32 without
+// re-reading HBM.
 template <typename T, int BLOCK>
 __global__ void layernorm_fwd_kernel(
     const T* __restrict__ x, const float* __restrict__ w,
@@ -28,32 +31,45 @@ __global__ void layernorm_fwd_kernel(
   const T* xr = x + row * (long)D;
   T* yr = y + row * (long)D;
 
-  // 8-wide vectorized accumulation (D assumed multiple of 8; model dims are)
-  float s = 0.f, s2 = 0.f;
-  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+  constexpr int MAX_CHUNKS = 4;  // 4 * 8 * BLOCK elems = 8192 at BLOCK=256
+  float xv[MAX_CHUNKS][8];
+  const int nchunks = (D + BLOCK * 8 - 1) / (BLOCK * 8);
+
+  float s = 0.f;
+  for (int c = 0; c < nchunks; ++c) {
+    const int i = (c * BLOCK + threadIdx.x) * 8;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float v = load_f32<T>(xr, i + j);
-      s += v;
-      s2 += v * v;
+      xv[c][j] = (i + j < D) ? load_f32<T>(xr, i + j) : 0.f;
+      s += xv[c][j];
     }
   }
   s = block_reduce_sum(s, scratch);
-  s2 = block_reduce_sum(s2, scratch);
   const float mean = s / D;
-  const float var = fmaxf(s2 / D - mean * mean, 0.f);
-  const float rstd = rsqrtf(var + eps);
+  float s2 = 0.f;
+  for (int c = 0; c < nchunks; ++c) {
+    const int i = (c * BLOCK + threadIdx.x) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float d = (i + j < D) ? (xv[c][j] - mean) : 0.f;
+      s2 += d * d;
+    }
+  }
+  s2 = block_reduce_sum(s2, scratch);
+  const float rstd = rsqrtf(s2 / D + eps);
   if (threadIdx.x == 0) {
     mean_out[row] = mean;
     rstd_out[row] = rstd;
   }
-  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+  for (int c = 0; c < nchunks; ++c) {
+    const int i = (c * BLOCK + threadIdx.x) * 8;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float v = load_f32<T>(xr, i + j);
-      float wj = w ? w[i + j] : 1.f;
-      float bj = b ? b[i + j] : 0.f;
-      store_f32<T>(yr, i + j, (v - mean) * rstd * wj + bj);
+      if (i + j < D) {
+        const float wj = w ? w[i + j] : 1.f;
+        const float bj = b ? b[i + j] : 0.f;
+        store_f32<T>(yr, i + j, (xv[c][j] - mean) * rstd * wj + bj);
+      }
     }
   }
 }

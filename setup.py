@@ -24,6 +24,7 @@ sources = [
     str(HIP_DIR / "cross_entropy.hip"),
     str(HIP_DIR / "optim.hip"),
     str(HIP_DIR / "attention.hip"),
+    str(HIP_DIR / "debug.hip"),
 ]
 
 setup(

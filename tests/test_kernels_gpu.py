@@ -43,8 +43,10 @@ def test_layernorm_fwd(dev, ext, D, dtype):
     b = torch.randn(D, device=dev) * 0.1
     y, mean, rstd = ext.layernorm_fwd(x, w, b, 1e-5)
     ref = torch.nn.functional.layer_norm(x.float(), (D,), w, b, 1e-5)
-    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
-    assert (y.float() - ref).abs().max() < tol
+    # bf16 output-store rounding is ~2^-8 relative; compare relative to |ref|
+    tol = 1e-2 if dtype == torch.bfloat16 else 1e-5
+    denom = float(ref.abs().max().clamp_min(1.0))
+    assert (y.float() - ref).abs().max() / denom < tol
     assert (mean - x.float().mean(-1)).abs().max() < 1e-4
 
 
