@@ -42,7 +42,8 @@ class FusedLayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(normalized_shape)) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if use_hip(x):
+        d = self.normalized_shape[0]
+        if d % 256 == 0 and d <= 4096 and use_hip(x):
             return _LayerNormHIP.apply(
                 x.contiguous(), self.weight, self.bias, self.eps
             )

@@ -90,15 +90,23 @@ class Trainer:
     def train_batch(self, batches: list[dict]) -> float:
         """One optimization batch = grad_accum microbatches. Returns loss."""
         self.optimizer.zero_grad(set_to_none=True)
-        total_loss = 0.0
         n = len(batches)
-        for mb in batches:
-            ids = mb["input_ids"].to(self.device, non_blocking=True)
-            with self.autocast():
+        total_loss_t = None
+        # ONE autocast region for the whole accumulation loop: the autocast
+        # weight-cast cache then converts each fp32 weight to bf16 once per
+        # optimization batch instead of once per microbatch (measured ~8% of
+        # step time in cast/copy kernels at grad_accum=16, profiles/r01).
+        with self.autocast():
+            for mb in batches:
+                ids = mb["input_ids"].to(self.device, non_blocking=True)
                 out = self.model(ids, labels=ids)
-            loss = out["loss"] / n
-            loss.backward()
-            total_loss += float(loss.detach())
+                loss = out["loss"] / n
+                loss.backward()
+                total_loss_t = (
+                    loss.detach()
+                    if total_loss_t is None
+                    else total_loss_t + loss.detach()
+                )
         if self.grad_sync_hook is not None:
             self.grad_sync_hook(self.model)
         if self.clip_norm > 0:
@@ -108,7 +116,9 @@ class Trainer:
         samples = sum(b["input_ids"].shape[0] for b in batches) * self.world_size
         tokens = samples * batches[0]["input_ids"].shape[1]
         self.timestamp.tick_batch(samples, tokens)
-        return total_loss
+        # Single device->host sync per optimization batch, after all kernels
+        # for the step are enqueued.
+        return float(total_loss_t) if total_loss_t is not None else 0.0
 
     def fit(self, duration_batches: int | str, callback=None) -> dict:
         """Train for `duration_batches` optimization batches; returns metrics."""
