@@ -1,0 +1,701 @@
+// Flash attention (causal, fused ALiBi) for CDNA4 gfx950 — MFMA bf16.
+//
+// Replaces flash-attn 2.6.3's CUDA kernels (reference install_env.sh:71;
+// SURVEY.md §2.3 rows 1-2). MI355X-native structure (not a port):
+//
+//   * 256-thread workgroups = 4 wave64s; each wave owns a 32-row Q block
+//     (fwd / dQ) or 32 keys (dK/dV); mfma_f32_32x32x16_bf16 tiles.
+//   * "Swapped" QK^T — mfma(A=K, B=Q) gives S[key][q] with q = lane&31, so
+//     the online-softmax state (m, l) is lane-local: rescales are scalar
+//     per lane, row reductions are 16 regs + one shfl_xor(32).
+//   * P (f32 regs) is converted to the next MFMA's B-operand fragments
+//     in-register with v_cvt_pk_bf16_f32 + permlane32_swap — no LDS
+//     round-trip for P.
+//   * K/V tiles staged in LDS: row image [32][D] with a ((row&15)<<4) XOR
+//     byte swizzle (conflict-free b128 column reads) + an explicitly
+//     transposed image [D][32] for the A-operands of PV / dQ / dK / dV
+//     (upgrade path: ds_read_b64_tr_b16 hardware transpose reads).
+//   * Online softmax carries LSE out for the backward; backward is the
+//     standard FlashAttention-2 split: one kernel for dK/dV (blocks own key
+//     tiles, loop over Q) and one for dQ (blocks own Q tiles, loop over
+//     KV) — atomics-free and bit-deterministic.
+//
+// v1 is correctness-first: single-buffered LDS, one barrier pair per tile.
+
+#pragma once
+#include "common.h"
+
+namespace photon_hip {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+constexpr int ATT_BLOCK = 256;   // 4 waves
+constexpr int QB = 32;           // q rows per wave
+constexpr int KB = 32;           // keys per kv tile
+constexpr int WAVES = 4;
+
+DEV_INLINE unsigned swz(unsigned byte, int row) {
+  return byte ^ (((unsigned)row & 15u) << 4);
+}
+
+DEV_INLINE unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// Pack 8 consecutive f32 P-regs (reg base rb) into the MFMA B-operand
+// fragment for one 16-deep k-step. Derivation: reg r holds row index
+// (r&3)+8*(r>>2)+4*hi; the B fragment wants rows 8*hi+jj. cvt_pk pairs +
+// permlane32_swap (upper(vdst) <-> lower(src)) rearrange exactly that.
+DEV_INLINE bf16x8 pack_bfrag(const float* p, int rb) {
+  unsigned d0 = cvt_pk_bf16(p[rb + 0], p[rb + 1]);
+  unsigned d1 = cvt_pk_bf16(p[rb + 2], p[rb + 3]);
+  unsigned d2 = cvt_pk_bf16(p[rb + 4], p[rb + 5]);
+  unsigned d3 = cvt_pk_bf16(p[rb + 6], p[rb + 7]);
+  auto r02 = __builtin_amdgcn_permlane32_swap(d0, d2, false, false);
+  auto r13 = __builtin_amdgcn_permlane32_swap(d1, d3, false, false);
+  union {
+    unsigned u[4];
+    bf16x8 v;
+  } out;
+  out.u[0] = r02[0];
+  out.u[1] = r13[0];
+  out.u[2] = r02[1];
+  out.u[3] = r13[1];
+  return out.v;
+}
+
+// Load a 16-byte (8 x bf16) fragment from an LDS row image with the XOR
+// swizzle. byte = row*rowstride + coloff must be 16B-aligned pre-swizzle.
+DEV_INLINE bf16x8 lds_frag(const __bf16* img, int row, int rowstride_b,
+                           int coloff_b) {
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  return *(const bf16x8*)((const char*)img + byte);
+}
+
+DEV_INLINE void lds_store16(__bf16* img, int row, int rowstride_b,
+                            int coloff_b, bf16x8 v) {
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  *(bf16x8*)((char*)img + byte) = v;
+}
+
+DEV_INLINE void lds_store2(__bf16* img, int row, int rowstride_b,
+                           int coloff_b, __bf16 v) {
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  *(__bf16*)((char*)img + byte) = v;
+}
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// Hardware transpose read (gfx950 ds_read_b64_tr_b16). Probed semantics
+// (scripts/probe_tr.hip, addr_mode 3): per 16-lane group,
+// out[j][elem k] = element (j&3) of the 8-byte chunk loaded by lane
+// (4k + (j>>2)). With lane i pointing at row (i>>2), 4-elem column chunk
+// 4*(i&3) of a row-major image, output lane j receives column j of rows
+// 0..3 in NATURAL order. Used to build V^T / K^T / Q^T MFMA A-fragments
+// straight from ROW-major LDS images (no scalar-store transposed image —
+// that image's stores were 4-way (D=64) to 32-way (D=128) bank-conflicted).
+DEV_INLINE bf16x4 lds_tr16(const __bf16* img, unsigned byte) {
+  typedef __attribute__((address_space(3))) bf16x4* lds_v4p;
+  typedef __attribute__((address_space(3))) char* lds_cp;
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_v4p)((lds_cp)img + byte));
+}
+
+// Stage a [rows=32][D] global tile into (a) the row image (swizzled,
+// row stride D*2 bytes) and (b) the transposed image [D][32] (swizzled,
+// row stride 64 bytes). Cooperative across the whole 256-thread block.
+// Rows >= rows_valid are zero-filled.
+template <int D>
+DEV_INLINE void stage_tile(const __bf16* __restrict__ gsrc, long g_row0,
+                           long g_rows_total, long g_row_stride,
+                           __bf16* row_img, __bf16* t_img) {
+  constexpr int CHUNKS = 32 * D / 8;  // 16B chunks
+  for (int c = threadIdx.x; c < CHUNKS; c += ATT_BLOCK) {
+    const int row = c / (D / 8);
+    const int col = (c % (D / 8)) * 8;
+    bf16x8 v;
+    const long grow = g_row0 + row;
+    if (grow < g_rows_total) {
+      v = *(const bf16x8*)(gsrc + grow * g_row_stride + col);
+    } else {
+      v = bf16x8{};
+    }
+    if (row_img) lds_store16(row_img, row, D * 2, col * 2, v);
+#ifndef ABENCH_NO_VT
+    if (t_img) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        lds_store2(t_img, col + j, 64, row * 2, v[j]);
+      }
+    }
+#endif
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Forward
+//
+// v2 structure (each measured on MI355X, see profiles/):
+//   * KV tile = 64 keys (KBF), double-buffered LDS, ONE barrier per tile.
+//   * Register-staged loads issued at the top of the compute phase for the
+//     NEXT tile (T14 async-stage: HBM latency hides under this tile's
+//     MFMAs), ds_writes land after the compute, into the other buffer.
+//   * Defer-max online softmax (T13, THR=8): o_acc rescale only when a
+//     lane's tile max outgrows the running max by more than THR; P is then
+//     bounded by e^8 in fp32 accumulate (accuracy cost ~3x vs THR=0,
+//     validated against fp64 + spiked-key data in tests).
+//   * Mask fast-path: tiles strictly below the causal diagonal skip the
+//     mask compare per element.
+// ---------------------------------------------------------------------------
+constexpr int KBF = 64;  // fwd kv-tile keys
+
+#ifndef ATT_FWD_MINWAVES
+#define ATT_FWD_MINWAVES 2  // min waves/SIMD: caps register alloc at 256
+#endif
+
+template <int D>
+__global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
+    const __bf16* __restrict__ q, const __bf16* __restrict__ k,
+    const __bf16* __restrict__ v, const float* __restrict__ slopes,
+    __bf16* __restrict__ out, float* __restrict__ lse_out, int S, int H,
+    int causal) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // Per buffer: k row image [KBF][D] + v transposed image [D][KBF].
+  constexpr int IMG = KBF * D * 2;  // bytes per image
+  // Buffer b: k at smem + b*2*IMG, v^T at smem + b*2*IMG + IMG.
+  auto k_img = [&](int b) { return (__bf16*)(smem + b * 2 * IMG); };
+  auto v_img = [&](int b) { return (__bf16*)(smem + b * 2 * IMG + IMG); };
+  // epilogue bounce reuses smem from offset 0: [wave][32][D]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const long bh = blockIdx.y;
+  const int h = bh % H;
+  const float slope = slopes[h];
+  const float scale = rsqrtf((float)D);
+
+  const long base = bh * (long)S * D;
+  const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
+  const int my_q = q0 + lq;  // this lane's q row
+
+  // Q fragments in registers: B-operand, frag kk covers dh [kk*16, kk*16+16)
+  bf16x8 qfrag[D / 16];
+  {
+    const long qrow = base + (long)min(my_q, S - 1) * D;
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      if (my_q < S) {
+        qfrag[kk] = *(const bf16x8*)(q + qrow + kk * 16 + 8 * hi);
+      } else {
+        qfrag[kk] = bf16x8{};
+      }
+    }
+  }
+
+  float m_run = -1e30f;  // finite sentinel: exp(-huge) == 0, no NaN paths
+  float l_run = 0.f;
+  f32x16 o_acc[D / 32];
+#pragma unroll
+  for (int db = 0; db < D / 32; ++db) o_acc[db] = f32x16{};
+
+  const int q_max_block =
+      min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
+  const int n_tiles = causal ? (q_max_block / KBF + 1) : ((S + KBF - 1) / KBF);
+  const int my_q_max = min(q0 + QB - 1, S - 1);
+
+  // Register staging: each thread owns KBF*D/8/ATT_BLOCK 16-byte chunks per
+  // tensor. Chunk c -> key row c / (D/8), col (c % (D/8)) * 8.
+  constexpr int NCHUNK = KBF * D / 8 / ATT_BLOCK;
+  bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
+
+  auto stage_load = [&](int t) {
+#pragma unroll
+    for (int i = 0; i < NCHUNK; ++i) {
+      const int c = i * ATT_BLOCK + threadIdx.x;
+      const int row = c / (D / 8);
+      const int col = (c % (D / 8)) * 8;
+      const long grow = (long)t * KBF + row;
+#ifndef ABENCH_NO_LOAD
+      if (grow < S) {
+        k_stage[i] = *(const bf16x8*)(k + base + grow * D + col);
+        v_stage[i] = *(const bf16x8*)(v + base + grow * D + col);
+      } else {
+        k_stage[i] = bf16x8{};
+        v_stage[i] = bf16x8{};
+      }
+#else
+      k_stage[i] = bf16x8{};
+      v_stage[i] = bf16x8{};
+      (void)grow;
+#endif
+    }
+  };
+  auto stage_write = [&](int b) {
+#pragma unroll
+    for (int i = 0; i < NCHUNK; ++i) {
+      const int c = i * ATT_BLOCK + threadIdx.x;
+      const int row = c / (D / 8);
+      const int col = (c % (D / 8)) * 8;
+#ifndef ABENCH_NO_KSTORE
+      lds_store16(k_img(b), row, D * 2, col * 2, k_stage[i]);
+#endif
+#ifndef ABENCH_NO_VT
+      // V is a ROW image too; PV reads it transposed via lds_tr16.
+      lds_store16(v_img(b), row, D * 2, col * 2, v_stage[i]);
+#endif
+    }
+  };
+
+  stage_load(0);
+  stage_write(0);
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int buf = t & 1;
+    const int kv0 = t * KBF;
+    if (t + 1 < n_tiles) stage_load(t + 1);  // hide HBM under this tile
+#ifndef ABENCH_NO_BARRIER
+    __syncthreads();  // LDS[buf] writes (prev iter) visible to all waves
+#endif
+
+    const bool active = !causal || (kv0 <= my_q_max);
+    if (active) {
+      // Two 32-key SUBTILES over the staged 64-key buffer: halves the live
+      // softmax register state (s_acc + p = 32 regs instead of 64) so the
+      // MFMA accumulators stay in arch VGPRs — the v1 64-key softmax spent
+      // more issue slots on v_accvgpr shuttling than on MFMAs (pmc1
+      // profile: MFMA was 13% of issued instruction time).
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int kv0s = kv0 + sub * 32;
+        if (causal && kv0s > my_q_max) break;
+        // S[key][q] = K Q^T : A = K row frags, B = Q regs.
+        f32x16 s_acc = f32x16{};
+#ifndef ABENCH_NO_QK
+#pragma unroll
+        for (int kk = 0; kk < D / 16; ++kk) {
+          bf16x8 a =
+              lds_frag(k_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qfrag[kk], s_acc,
+                                                          0, 0, 0);
+        }
+#endif
+        // sv = s*scale + slope*(key - my_q); masked lanes get -1e30 so the
+        // exp pass needs no per-element compare (exp(-huge - m) == 0).
+        // key = kv0s + PAT(r), PAT compile-time, so ALiBi is 2 fma/element.
+        // need_mask is BLOCK-uniform (scalar branch): a wave-dependent
+        // condition here gets if-converted into per-element cmp+cndmask on
+        // every subtile (measured 0.65 ms of the 1.2 ms kernel).
+        const bool need_mask =
+            (kv0s + 32 > S) ||
+            (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
+        const float abase = slope * (float)(kv0s - my_q);
+        float p[16];
+        float tile_max = -1e30f;
+#ifdef ABENCH_NO_SOFTMAX
+#pragma unroll
+        for (int r = 0; r < 16; ++r) p[r] = s_acc[r];
+        (void)need_mask; (void)abase; (void)tile_max;
+#else
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
+          if (need_mask) {
+            const int key = kv0s + pat;
+            const bool masked = (key >= S) || (causal && key > my_q);
+            sv = masked ? -1e30f : sv;
+          }
+          p[r] = sv;
+          tile_max = fmaxf(tile_max, sv);
+        }
+        tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+        // T13 defer-max: rescale only if some lane's max grew by > THR.
+        constexpr float THR = 8.f;
+        if (!__all(tile_max - m_run <= THR)) {
+          const float m_new = fmaxf(m_run, tile_max);
+          const float alpha = __expf(m_run - m_new);  // exp(-huge) == 0
+          l_run *= alpha;
+#pragma unroll
+          for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+            for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+          }
+          m_run = m_new;
+        }
+        float l_add = 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+#ifndef ABENCH_NO_EXP
+          p[r] = __expf(p[r] - m_run);  // masked: exp(-huge) == 0
+#else
+          p[r] = p[r] - m_run;  // bench-only: perf bisect without v_exp
+#endif
+          l_add += p[r];
+        }
+        l_add += __shfl_xor(l_add, 32, 64);
+        l_run += l_add;
+#endif  // ABENCH_NO_SOFTMAX
+        // PV: O2[pi(dh)][q] += V^T P ; A-frags built by hardware
+        // transpose reads from the V ROW image. The A rows come out
+        // permuted within each 16-row half (pi: row 16*g + j holds
+        // dh 16*g + 4*(j&3) + (j>>2)); the epilogue un-permutes.
+#ifndef ABENCH_NO_PV
+        const int tj = lane & 15;         // position within 16-lane group
+        const int tg1 = (lane >> 4) & 1;  // dh half of the 32-row frag
+#pragma unroll
+        for (int s16 = 0; s16 < 2; ++s16) {
+          bf16x8 pfrag = pack_bfrag(p, 8 * s16);
+          const int koff = sub * 32 + s16 * 16 + 8 * hi + (tj >> 2);
+#pragma unroll
+          for (int db = 0; db < D / 32; ++db) {
+            const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
+            union { bf16x4 h[2]; bf16x8 v8; } a;
+#pragma unroll
+            for (int rd = 0; rd < 2; ++rd) {
+              const int key = koff + 4 * rd;
+              a.h[rd] = lds_tr16(
+                  v_img(buf), swz((unsigned)(key * (D * 2) + dhc * 2), key));
+            }
+            o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                a.v8, pfrag, o_acc[db], 0, 0, 0);
+          }
+        }
+#endif
+      }
+    }
+    if (t + 1 < n_tiles) stage_write(1 - buf);
+  }
+  __syncthreads();  // protect epilogue smem reuse
+
+  // epilogue: bounce O through LDS for coalesced stores. The tr-read
+  // addressing (probed: out[j][k] = elem (j&3) of lane (4k + (j>>2)))
+  // delivers A-frag rows in NATURAL order — no permutation to undo.
+  __bf16* o_img = (__bf16*)smem + wave * 32 * D;  // per-wave [32][D], linear
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+  for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int dh = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      o_img[lq * D + dh] = (__bf16)(o_acc[db][r] * inv_l);
+    }
+  }
+  if (hi == 0 && my_q < S && lse_out) {
+    lse_out[bh * (long)S + my_q] = m_run + __logf(l_run);
+  }
+  __syncthreads();
+  // each wave stores its own 32 rows
+  for (int c = lane; c < 32 * D / 8; c += 64) {
+    const int row = c / (D / 8);
+    const int col = (c % (D / 8)) * 8;
+    if (q0 + row < S) {
+      *(bf16x8*)(out + base + (long)(q0 + row) * D + col) =
+          *(const bf16x8*)(o_img + row * D + col);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward dQ: blocks own Q tiles, loop over KV tiles.
+// dQ^T[dh][q] = sum_key K^T[dh][key] * dS[key][q] * scale
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
+    const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
+    const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    const float* __restrict__ slopes, const float* __restrict__ lse,
+    const float* __restrict__ delta, __bf16* __restrict__ dq, int S, int H,
+    int causal) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* k_img = (__bf16*)smem;             // [32][D] row image
+  __bf16* v_img = (__bf16*)(smem + 64 * D);  // [32][D] row image
+  // K^T A-fragments come from k_img via lds_tr16 (no transposed image).
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const long bh = blockIdx.y;
+  const int h = bh % H;
+  const float slope = slopes[h];
+  const float scale = rsqrtf((float)D);
+  const long base = bh * (long)S * D;
+  const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
+  const int my_q = q0 + lq;
+
+  bf16x8 qfrag[D / 16], dofrag[D / 16];
+  {
+    const long row = base + (long)min(my_q, S - 1) * D;
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      if (my_q < S) {
+        qfrag[kk] = *(const bf16x8*)(q + row + kk * 16 + 8 * hi);
+        dofrag[kk] = *(const bf16x8*)(dout + row + kk * 16 + 8 * hi);
+      } else {
+        qfrag[kk] = bf16x8{};
+        dofrag[kk] = bf16x8{};
+      }
+    }
+  }
+  const float my_lse = (my_q < S) ? lse[bh * (long)S + my_q] : INFINITY;
+  const float my_delta = (my_q < S) ? delta[bh * (long)S + my_q] : 0.f;
+
+  f32x16 dq_acc[D / 32];
+#pragma unroll
+  for (int db = 0; db < D / 32; ++db) dq_acc[db] = f32x16{};
+
+  const int q_max_block = min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
+  const int n_tiles = causal ? (q_max_block / KB + 1) : ((S + KB - 1) / KB);
+  const int my_q_max = min(q0 + QB - 1, S - 1);
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * KB;
+    stage_tile<D>(k + base, kv0, S, D, k_img, nullptr);
+    stage_tile<D>(v + base, kv0, S, D, v_img, nullptr);
+    __syncthreads();
+
+    const bool active = !causal || (kv0 <= my_q_max);
+    if (active) {
+      f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
+#pragma unroll
+      for (int kk = 0; kk < D / 16; ++kk) {
+        bf16x8 ka = lds_frag(k_img, lq, D * 2, kk * 32 + hi * 16);
+        bf16x8 va = lds_frag(v_img, lq, D * 2, kk * 32 + hi * 16);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[kk], s_acc,
+                                                        0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[kk],
+                                                         dp_acc, 0, 0, 0);
+      }
+      // p = exp(sv - lse); masked keys land at exp(-huge) == 0 without a
+      // per-element compare (finite sentinel; cf. fwd).
+      const bool need_mask =
+          (kv0 + KB > S) ||
+          (causal && (kv0 + KB - 1 > (int)blockIdx.x * (WAVES * QB)));
+      const float abase = slope * (float)(kv0 - my_q);
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
+        if (need_mask) {
+          const int key = kv0 + pat;
+          const bool masked = (key >= S) || (causal && key > my_q);
+          sv = masked ? -1e30f : sv;
+        }
+        const float pv = __expf(sv - my_lse);
+        ds[r] = pv * (dp_acc[r] - my_delta) * scale;
+      }
+      const int tj = lane & 15;
+      const int tg1 = (lane >> 4) & 1;
+#pragma unroll
+      for (int s16 = 0; s16 < 2; ++s16) {
+        bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
+        const int koff = s16 * 16 + 8 * hi + (tj >> 2);
+#pragma unroll
+        for (int db = 0; db < D / 32; ++db) {
+          const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
+          union { bf16x4 h[2]; bf16x8 v8; } a;
+#pragma unroll
+          for (int rd = 0; rd < 2; ++rd) {
+            const int key = koff + 4 * rd;
+            a.h[rd] = lds_tr16(
+                k_img, swz((unsigned)(key * (D * 2) + dhc * 2), key));
+          }
+          dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a.v8, dsfrag, dq_acc[db], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue (tr-read rows are natural order; see fwd epilogue note)
+  __bf16* o_img = (__bf16*)smem + wave * 32 * D;
+#pragma unroll
+  for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int dh = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      o_img[lq * D + dh] = (__bf16)dq_acc[db][r];
+    }
+  }
+  __syncthreads();
+  for (int c = lane; c < 32 * D / 8; c += 64) {
+    const int row = c / (D / 8);
+    const int col = (c % (D / 8)) * 8;
+    if (q0 + row < S) {
+      *(bf16x8*)(dq + base + (long)(q0 + row) * D + col) =
+          *(const bf16x8*)(o_img + row * D + col);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward dK/dV: blocks own key tiles, loop over Q tiles.
+//   dV^T[dh][key] = sum_q dO^T[dh][q] P[q][key]
+//   dK^T[dh][key] = sum_q Q^T[dh][q] dS[q][key] * scale
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dkdv_kernel(
+    const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
+    const __bf16* __restrict__ k, const __bf16* __restrict__ v,
+    const float* __restrict__ slopes, const float* __restrict__ lse,
+    const float* __restrict__ delta, __bf16* __restrict__ dk,
+    __bf16* __restrict__ dv, int S, int H, int causal) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* q_img = (__bf16*)smem;              // [32][D] row image
+  __bf16* do_img = (__bf16*)(smem + 64 * D);  // [32][D] row image
+  // Q^T / dO^T A-fragments come via lds_tr16 (no transposed images).
+  float* lse_t = (float*)(smem + 128 * D);    // [32]
+  float* del_t = lse_t + 32;                  // [32]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const long bh = blockIdx.y;
+  const int h = bh % H;
+  const float slope = slopes[h];
+  const float scale = rsqrtf((float)D);
+  const long base = bh * (long)S * D;
+  const int k0 = blockIdx.x * (WAVES * KB) + wave * KB;
+  const int my_key = k0 + lq;
+
+  // K, V rows of this wave's keys as B-operand fragments (like Q in fwd)
+  bf16x8 kfrag[D / 16], vfrag[D / 16];
+  {
+    const long row = base + (long)min(my_key, S - 1) * D;
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      if (my_key < S) {
+        kfrag[kk] = *(const bf16x8*)(k + row + kk * 16 + 8 * hi);
+        vfrag[kk] = *(const bf16x8*)(v + row + kk * 16 + 8 * hi);
+      } else {
+        kfrag[kk] = bf16x8{};
+        vfrag[kk] = bf16x8{};
+      }
+    }
+  }
+
+  f32x16 dk_acc[D / 32], dv_acc[D / 32];
+#pragma unroll
+  for (int db = 0; db < D / 32; ++db) {
+    dk_acc[db] = f32x16{};
+    dv_acc[db] = f32x16{};
+  }
+
+  const int k_min_block = blockIdx.x * (WAVES * KB);
+  const int t0 = causal ? (k_min_block / QB) : 0;
+  const int n_tiles = (S + QB - 1) / QB;
+  const int my_k_min = k0;
+
+  for (int t = t0; t < n_tiles; ++t) {
+    const int qt0 = t * QB;
+    stage_tile<D>(q + base, qt0, S, D, q_img, nullptr);
+    stage_tile<D>(dout + base, qt0, S, D, do_img, nullptr);
+    for (int i = threadIdx.x; i < 32; i += ATT_BLOCK) {
+      const int qi = qt0 + i;
+      lse_t[i] = (qi < S) ? lse[bh * (long)S + qi] : INFINITY;
+      del_t[i] = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
+    }
+    __syncthreads();
+
+    const bool active = !causal || (qt0 + QB - 1 >= my_k_min);
+    if (active) {
+      // S'[q][key]: A = Q row frags, B = K regs; dP'[q][key]: A = dO, B = V
+      f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
+#pragma unroll
+      for (int kk = 0; kk < D / 16; ++kk) {
+        bf16x8 qa = lds_frag(q_img, lq, D * 2, kk * 32 + hi * 16);
+        bf16x8 doa = lds_frag(do_img, lq, D * 2, kk * 32 + hi * 16);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[kk], s_acc,
+                                                        0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vfrag[kk],
+                                                         dp_acc, 0, 0, 0);
+      }
+      // ALiBi: bias = slope*(my_key - qi) = fma(slope, -PAT, abase);
+      // masked q rows land at exp(-huge - l) == 0 (finite sentinel).
+      // Diagonal-block detection is BLOCK-uniform (scalar branch).
+      const bool need_mask =
+          (qt0 + QB > S) ||
+          (causal && (int)(blockIdx.x + 1) * (WAVES * KB) - 1 >= qt0);
+      const float abase = slope * (float)(my_key - qt0);
+      float p[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float l = lse_t[pat];
+        const float dlt = del_t[pat];
+        float sv = fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
+        if (need_mask) {
+          const int qi = qt0 + pat;
+          const bool masked =
+              (my_key >= S) || (causal && my_key > qi) || (qi >= S);
+          sv = masked ? -1e30f : sv;
+        }
+        p[r] = __expf(sv - l);
+        ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
+      }
+      const int tj = lane & 15;
+      const int tg1 = (lane >> 4) & 1;
+#pragma unroll
+      for (int s16 = 0; s16 < 2; ++s16) {
+        bf16x8 pfrag = pack_bfrag(p, 8 * s16);
+        bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
+        const int qoff = s16 * 16 + 8 * hi + (tj >> 2);
+#pragma unroll
+        for (int db = 0; db < D / 32; ++db) {
+          const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
+          union { bf16x4 h[2]; bf16x8 v8; } doa, qa;
+#pragma unroll
+          for (int rd = 0; rd < 2; ++rd) {
+            const int qr = qoff + 4 * rd;
+            const unsigned byte =
+                swz((unsigned)(qr * (D * 2) + dhc * 2), qr);
+            doa.h[rd] = lds_tr16(do_img, byte);
+            qa.h[rd] = lds_tr16(q_img, byte);
+          }
+          dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              doa.v8, pfrag, dv_acc[db], 0, 0, 0);
+          dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              qa.v8, dsfrag, dk_acc[db], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: two bounces (dk then dv) through per-wave LDS
+  // (tr-read rows are natural order; see fwd epilogue note)
+  __bf16* o_img = (__bf16*)smem + wave * 32 * D;
+  for (int which = 0; which < 2; ++which) {
+    f32x16* acc = which == 0 ? dk_acc : dv_acc;
+    __bf16* dst = which == 0 ? dk : dv;
+#pragma unroll
+    for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int dh = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        o_img[lq * D + dh] = (__bf16)acc[db][r];
+      }
+    }
+    __syncthreads();
+    for (int c = lane; c < 32 * D / 8; c += 64) {
+      const int row = c / (D / 8);
+      const int col = (c % (D / 8)) * 8;
+      if (k0 + row < S) {
+        *(bf16x8*)(dst + base + (long)(k0 + row) * D + col) =
+            *(const bf16x8*)(o_img + row * D + col);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+
+}  // namespace photon_hip

@@ -18,6 +18,14 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = Path(__file__).parent
 HIP_DIR = ROOT / "photon_amd" / "ops" / "hip"
 
+# torch's hipify caches *_hip.hip / *_hip.h copies and does NOT re-hipify
+# when only an included header changed — which silently builds stale device
+# code (cost us a debugging session). Always clear the generated copies.
+for stale in HIP_DIR.glob("*_hip.hip"):
+    stale.unlink()
+for stale in HIP_DIR.glob("*_hip.h"):
+    stale.unlink()
+
 sources = [
     str(HIP_DIR / "bindings.cpp"),
     str(HIP_DIR / "layernorm.hip"),
