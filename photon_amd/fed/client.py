@@ -15,6 +15,11 @@ from ..data import build_eval_loader, build_train_loader
 from ..models import build_model
 from ..train import Trainer
 from .flat import FlatParams
+from .params_ops import (
+    manipulate_pre_training,
+    post_process_client_result,
+    set_optimizer_state,
+)
 
 
 @dataclass
@@ -46,6 +51,8 @@ class FedClient:
         # client runs on its own rank or shares one.
         self._timestamps: dict[int, dict] = {}
         self._loader_states: dict[int, dict] = {}
+        # last local params per cid, kept only when personalized_layers is on
+        self._personal: dict[int, torch.Tensor] = {}
 
     def _ensure_trainer(self, cid: int) -> Trainer:
         llm = self.cfg["llm_config"]
@@ -68,19 +75,22 @@ class FedClient:
     def fit(
         self,
         cid: int,
-        global_flat: torch.Tensor,
+        payload: torch.Tensor,
         layout: FlatParams,
         server_round: int,
         local_steps=None,
         reset_optimizer: bool = True,
     ) -> tuple[torch.Tensor, float, dict]:
-        """Run local_steps batches from the global params.
+        """Run local_steps batches from the incoming payload.
 
-        Returns (local flat params fp32, n_samples, metrics). n_samples =
-        local_steps * global_train_batch_size (reference
-        post_process_client_result, clients/utils.py:514-652).
+        payload is the flat global params, or [params | m1 | m2] when
+        fl.aggregate_momenta. Returns (outgoing payload fp32, n_samples,
+        metrics); n_samples = local_steps * global_train_batch_size
+        (reference post_process_client_result, clients/utils.py:514-652).
         """
         llm = self.cfg["llm_config"]
+        fl = self.cfg.get("fl", {})
+        momenta = bool(fl.get("aggregate_momenta", False))
         steps = duration_to_batches(
             local_steps if local_steps is not None else llm.get("local_steps", "500ba")
         )
@@ -91,9 +101,13 @@ class FedClient:
         trainer.timestamp = Timestamp()
         if cid in self._timestamps:
             trainer.timestamp.load_state_dict(self._timestamps[cid])
-        if cid in self._loader_states and not self.cfg["fl"].get("reset_dataset_state", False):
+        if cid in self._loader_states and not fl.get("reset_dataset_state", False):
             trainer.train_loader.load_state_dict(self._loader_states[cid])
+
         t0 = time.time()
+        global_flat, m1_in, m2_in = manipulate_pre_training(
+            payload, layout, fl, cid, local_params=self._personal.get(cid)
+        )
         # set params from the global buffer (HBM->HBM copies, no host hop)
         views = layout.layer_views_of(global_flat)
         params = dict(self.model.named_parameters())
@@ -104,6 +118,11 @@ class FedClient:
 
         if reset_optimizer:
             trainer.optimizer.state.clear()
+        if momenta and m1_in is not None:
+            # import aggregated momenta + step for bias correction
+            st_prev = self.client_states.get(cid)
+            step = st_prev.steps_done if st_prev else self._max_steps_done()
+            set_optimizer_state(trainer, layout, m1_in, m2_in, step=step)
 
         t1 = time.time()
         fit_metrics = trainer.fit(steps)
@@ -123,14 +142,26 @@ class FedClient:
         st.steps_done += steps
         self._timestamps[cid] = trainer.timestamp.state_dict()
         self._loader_states[cid] = trainer.train_loader.state_dict()
+        if fl.get("personalized_layers"):
+            self._personal[cid] = local_flat.clone()
+
+        out_payload, pp_metrics = post_process_client_result(
+            layout, global_flat, local_flat, n_samples,
+            trainer=trainer, aggregate_momenta=momenta,
+            report_layer_norms=bool(fl.get("report_layer_norms", False)),
+        )
         metrics = {
             "client/fit_set_parameters_time": set_params_time,
             "client/fit_time": fit_time,
             "client/fit_get_parameters_time": get_params_time,
             "loss/train/total": fit_metrics.get("loss/train/total", float("nan")),
             "steps_done": st.steps_done,
+            **pp_metrics,
         }
-        return local_flat, n_samples, metrics
+        return out_payload, n_samples, metrics
+
+    def _max_steps_done(self) -> int:
+        return max((s.steps_done for s in self.client_states.values()), default=0)
 
     @torch.no_grad()
     def evaluate(

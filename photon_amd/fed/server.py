@@ -38,6 +38,7 @@ from .client import FedClient
 from .flat import FlatParams
 from .noise_scale import FedSimpleNoiseScale
 from .runtime import Comm, assign_clients_to_ranks, sample_clients
+from .params_ops import join_payload, split_payload
 from .server_ckpt import (
     interpret_resume_round,
     obtain_sorted_rounds,
@@ -105,6 +106,10 @@ class FedServer:
         )
         self.server_steps_cumulative = 0
         self.start_round = 1
+        # aggregated client momenta when fl.aggregate_momenta
+        self.aggregate_momenta = bool(fl.get("aggregate_momenta", False))
+        self.client_m1 = self.layout.like() if self.aggregate_momenta else None
+        self.client_m2 = self.layout.like() if self.aggregate_momenta else None
 
     # -- initialization / resume -------------------------------------------
     def initialize(self) -> None:
@@ -142,7 +147,8 @@ class FedServer:
         assignment = assign_clients_to_ranks(sampled, self.comm.world_size)
         my_cids = assignment[self.comm.rank]
 
-        local_sum = torch.zeros_like(self.strategy.params)
+        outgoing = join_payload(self.strategy.params, self.client_m1, self.client_m2)
+        local_sum = torch.zeros_like(outgoing)
         local_weight = 0.0
         failures = 0
         steps_done_max = 0
@@ -152,19 +158,20 @@ class FedServer:
         t_fit = time.time()
         for cid in my_cids:
             try:
-                local_flat, n_samples, metrics = self.client.fit(
+                local_payload, n_samples, metrics = self.client.fit(
                     cid,
-                    self.strategy.params,
+                    outgoing,
                     self.layout,
                     server_round,
-                    reset_optimizer=bool(fl.get("reset_optimizer", True)),
+                    reset_optimizer=bool(fl.get("reset_optimizer", True))
+                    and not self.aggregate_momenta,
                 )
-                local_sum.add_(local_flat, alpha=n_samples)
+                local_sum.add_(local_payload, alpha=n_samples)
                 local_weight += n_samples
                 steps_done_max = max(steps_done_max, int(metrics.get("steps_done", 0)))
                 fit_metrics = metrics
                 if self.noise_scale is not None:
-                    g = self.strategy.params - local_flat
+                    g = self.strategy.params - local_payload[: self.layout.total]
                     per_client_sq_norms.append(
                         (n_samples, float(torch.dot(g, g)))
                     )
@@ -182,12 +189,18 @@ class FedServer:
                 f"(accept_failures_cnt={self.accept_failures_cnt})"
             )
 
-        # ONE weighted all-reduce over xGMI
+        # ONE weighted all-reduce over xGMI (params and momenta together)
         t_agg = time.time()
-        fedavg_flat, total_weight = self.comm.weighted_average_(local_sum, local_weight)
+        fedavg_payload, total_weight = self.comm.weighted_average_(local_sum, local_weight)
         agg_time = time.time() - t_agg
         if total_weight == 0:
             raise TooManyFailuresError(f"round {server_round}: no successful clients")
+        fedavg_flat, m1_avg, m2_avg = split_payload(
+            fedavg_payload, self.layout.total, self.aggregate_momenta
+        )
+        if self.aggregate_momenta:
+            self.client_m1.copy_(m1_avg)
+            self.client_m2.copy_(m2_avg)
 
         # replicated server-opt update
         strat_metrics = self.strategy.update(
@@ -202,6 +215,8 @@ class FedServer:
         steps_max_global = max(
             self.comm.all_gather_scalars(float(steps_done_max))
         )
+        # steps_done is cumulative per client, so the global max IS the
+        # cumulative server step count (fit_utils.py:179-183 semantics).
         self.server_steps_cumulative = int(steps_max_global)
 
         round_metrics = {

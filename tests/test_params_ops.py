@@ -1,0 +1,205 @@
+"""Tests for fed/params_ops.py + fed/store.py: momenta payload round-trip,
+layer personalization/randomization, freezing, checkers, object store."""
+
+import pytest
+import torch
+
+from photon_amd.conf import compose, config_yaml_dir
+from photon_amd.fed.flat import FlatParams
+from photon_amd.fed.params_ops import (
+    freeze_blocks,
+    get_optimizer_momenta,
+    join_payload,
+    l2_norm,
+    layer_l2_norms,
+    manipulate_pre_training,
+    parameters_checker,
+    personalize_layers,
+    post_process_client_result,
+    randomize_layers,
+    set_optimizer_state,
+    split_payload,
+)
+from photon_amd.fed.store import LocalStore, get_store
+from photon_amd.models.mpt import MPTCausalLM, MPTConfig
+from photon_amd.train import Trainer
+
+
+def tiny_model():
+    torch.manual_seed(0)
+    return MPTCausalLM(
+        MPTConfig(d_model=64, n_heads=2, n_layers=2, max_seq_len=32,
+                  vocab_size=128, attn_impl="torch", loss_impl="torch")
+    )
+
+
+def tiny_trainer(model):
+    llm = {
+        "optimizer": {"name": "decoupled_adamw", "lr": 1e-3,
+                      "betas": [0.9, 0.95], "eps": 1e-8, "weight_decay": 0.0},
+        "scheduler": {"schedulers": {"lr": {"name": "cosine_with_warmup",
+                                            "t_warmup": "1ba"}}},
+        "max_duration": "100ba",
+        "precision": "fp32",
+        "device_train_microbatch_size": 2,
+        "global_train_batch_size": 2,
+    }
+    return Trainer(model, llm, device="cpu")
+
+
+def test_payload_join_split_roundtrip():
+    p = torch.randn(10)
+    m1 = torch.randn(10)
+    m2 = torch.randn(10)
+    pay = join_payload(p, m1, m2)
+    assert pay.numel() == 30
+    p2, a, b = split_payload(pay, 10, momenta=True)
+    assert torch.equal(p2, p) and torch.equal(a, m1) and torch.equal(b, m2)
+    p3, a3, b3 = split_payload(p, 10, momenta=False)
+    assert a3 is None and torch.equal(p3, p)
+
+
+def test_momenta_export_import_roundtrip():
+    model = tiny_model()
+    trainer = tiny_trainer(model)
+    layout = FlatParams(model)
+    # take one step so the optimizer has momenta
+    ids = torch.randint(0, 128, (2, 16))
+    out = model(ids, labels=ids)
+    out["loss"].backward()
+    trainer.optimizer.step()
+    m1, m2 = get_optimizer_momenta(trainer, layout)
+    assert float(m1.abs().sum()) > 0
+    # wipe and re-import
+    trainer.optimizer.state.clear()
+    set_optimizer_state(trainer, layout, m1, m2, step=7)
+    m1b, m2b = get_optimizer_momenta(trainer, layout)
+    assert torch.allclose(m1, m1b) and torch.allclose(m2, m2b)
+    params = dict(model.named_parameters())
+    some_p = params[layout.names[0]]
+    assert trainer.optimizer.state[some_p]["step"] == 7
+
+
+def test_personalize_and_randomize_layers():
+    model = tiny_model()
+    layout = FlatParams(model)
+    layout.copy_from_model(model)
+    incoming = layout.clone_flat()
+    local = incoming + 1.0
+    chosen = personalize_layers(layout, incoming, local, ["wte"])
+    assert chosen, "wte should match"
+    views = dict(zip(layout.names, layout.layer_views_of(incoming)))
+    lviews = dict(zip(layout.names, layout.layer_views_of(local)))
+    for n in layout.names:
+        if n in chosen:
+            assert torch.equal(views[n], lviews[n])
+        else:
+            assert not torch.equal(views[n], lviews[n])
+    flat2 = layout.clone_flat()
+    r1 = randomize_layers(layout, flat2, ["norm_1"], seed=3)
+    assert r1
+    flat3 = layout.clone_flat()
+    randomize_layers(layout, flat3, ["norm_1"], seed=3)
+    assert torch.equal(flat2, flat3), "same seed => same randomization"
+
+
+def test_freeze_blocks():
+    model = tiny_model()
+    touched = freeze_blocks(model, frozen=["blocks.0"], unfrozen=["blocks.0.norm_1"])
+    assert touched
+    for n, p in model.named_parameters():
+        if "blocks.0.norm_1" in n:
+            assert p.requires_grad
+        elif "blocks.0" in n:
+            assert not p.requires_grad
+    # restore
+    for p in model.parameters():
+        p.requires_grad_(True)
+
+
+def test_parameters_checker():
+    a = torch.randn(5)
+    parameters_checker(a, a.clone(), equal=True)
+    with pytest.raises(AssertionError):
+        parameters_checker(a, a + 1, equal=True)
+    with pytest.raises(AssertionError):
+        parameters_checker(a, a.clone(), equal=False)
+
+
+def test_post_process_momenta_payload():
+    model = tiny_model()
+    trainer = tiny_trainer(model)
+    layout = FlatParams(model)
+    layout.copy_from_model(model)
+    g = layout.clone_flat()
+    local = g - 0.1
+    payload, metrics = post_process_client_result(
+        layout, g, local, 4.0, trainer=trainer, aggregate_momenta=True,
+        report_layer_norms=True,
+    )
+    assert payload.numel() == 3 * layout.total
+    assert "l2_norm_pseudo_gradient_client" in metrics
+    assert abs(metrics["l2_norm_pseudo_gradient_client"] -
+               l2_norm(torch.full_like(g, 0.1))) < 1e-4
+    assert set(metrics["layer_pseudo_grad_norms"]) == set(layout.names)
+
+
+def test_manipulate_pre_training_momenta_and_random():
+    model = tiny_model()
+    layout = FlatParams(model)
+    layout.copy_from_model(model)
+    p = layout.clone_flat()
+    m1 = torch.ones_like(p)
+    m2 = 2 * torch.ones_like(p)
+    fl = {"aggregate_momenta": True, "random_layers": ["norm_f"], "seed": 5}
+    params, a, b = manipulate_pre_training(join_payload(p, m1, m2), layout, fl, cid=0)
+    assert torch.equal(a, m1) and torch.equal(b, m2)
+    assert not torch.equal(params, p), "random_layers must change the payload"
+
+
+def test_local_store_roundtrip(tmp_path):
+    store = LocalStore(tmp_path)
+    store.write_bytes("run/server/3/state.bin", b"hello")
+    assert store.exists("run/server/3/state.bin")
+    assert store.read_bytes("run/server/3/state.bin") == b"hello"
+    assert store.list("run") == ["run/server/3/state.bin"]
+    store.delete("run")
+    assert store.list("run") == []
+
+
+def test_get_store_local(tmp_path):
+    cfg = {"comm_stack": {"s3": False, "shm": True},
+           "photon": {"saving_path": str(tmp_path)}}
+    store = get_store(cfg)
+    assert isinstance(store, LocalStore)
+
+
+def test_fed_round_with_momenta(tmp_path):
+    """Two federated rounds with aggregate_momenta on a tiny model: the
+    second round must import the first round's aggregated momenta."""
+    from photon_amd.fed.runtime import Comm
+    from photon_amd.fed.server import FedServer
+
+    cfg = compose(config_yaml_dir(), "base", [
+        "llm_config=mpt-125m",
+    ])
+    cfg = cfg.to_plain() if hasattr(cfg, "to_plain") else cfg
+    llm = cfg["llm_config"]
+    llm["model"].update({"d_model": 64, "n_heads": 2, "n_layers": 2,
+                         "max_seq_len": 32, "vocab_size": 128})
+    llm["model"]["attn_config"]["attn_impl"] = "torch"
+    llm["global_train_batch_size"] = 2
+    llm["device_train_microbatch_size"] = 2
+    llm["local_steps"] = "2ba"
+    llm["precision"] = "fp32"
+    cfg["fl"].update({
+        "n_total_clients": 2, "n_clients_per_round": 2, "n_rounds": 2,
+        "aggregate_momenta": True, "reset_optimizer": False,
+        "eval_period": 0,
+    })
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    server = FedServer(cfg, Comm(0, 1), device="cpu")
+    server.initialize()
+    server.run_round(1)
+    assert float(server.client_m1.abs().sum()) > 0, "momenta aggregated"
+    server.run_round(2)
