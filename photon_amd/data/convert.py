@@ -1,0 +1,175 @@
+"""Dataset conversion CLI — text/HF corpus -> per-client token shards.
+
+The analogue of the reference's photon/dataset/convert_dataset_hf.py
+(HF -> MDS shards per client, 1_gram.json token-frequency map + tokenizer
+dir, :313-363). Output layout mirrors the reference's
+``client_{i}/{split}`` directories, but shards are the photon_amd token
+format (data/shards.py) instead of MDS:
+
+    out_root/client_{i}/{split}/shard_*.bin + index.json + 1_gram.json
+    out_root/tokenizer/      (saved tokenizer, when one was used)
+
+Sources (this environment has no network):
+  * ``--source text:<path>``      newline-delimited text / .jsonl ("text" field)
+  * ``--source hf_disk:<path>``   a datasets.load_from_disk dataset
+  * ``--source synthetic:<n>``    n Zipf-sampled documents (bench/data-free CI)
+
+Tokenizer: ``--tokenizer <local dir>`` (transformers AutoTokenizer), or the
+built-in byte tokenizer (vocab 256 + BOS/EOS) when none is given.
+
+Usage:
+    python -m photon_amd.data.convert --source synthetic:1000 \
+        --out /tmp/shards --num-clients 8 --concat-tokens 2048
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+from collections import Counter
+from pathlib import Path
+
+import numpy as np
+
+from .shards import TokenShardWriter
+
+BYTE_EOS = 256
+BYTE_VOCAB = 258  # 256 bytes + EOS + pad
+
+
+class ByteTokenizer:
+    """Dependency-free fallback: UTF-8 bytes + EOS (vocab 258)."""
+
+    vocab_size = BYTE_VOCAB
+    eos_token_id = BYTE_EOS
+
+    def encode(self, text: str) -> list[int]:
+        return list(text.encode("utf-8"))
+
+    def save_pretrained(self, path):
+        Path(path).mkdir(parents=True, exist_ok=True)
+        (Path(path) / "tokenizer_config.json").write_text(
+            json.dumps({"tokenizer_class": "photon_amd.ByteTokenizer",
+                        "vocab_size": BYTE_VOCAB})
+        )
+
+
+def load_tokenizer(spec: str | None):
+    if spec is None:
+        return ByteTokenizer()
+    from transformers import AutoTokenizer
+
+    return AutoTokenizer.from_pretrained(spec)
+
+
+def iter_documents(source: str, seed: int = 1337):
+    kind, _, arg = source.partition(":")
+    if kind == "text":
+        path = Path(arg)
+        with open(path) as f:
+            for line in f:
+                line = line.strip()
+                if not line:
+                    continue
+                if line.startswith("{"):
+                    try:
+                        yield json.loads(line).get("text", "")
+                        continue
+                    except json.JSONDecodeError:
+                        pass
+                yield line
+    elif kind == "hf_disk":
+        import datasets
+
+        ds = datasets.load_from_disk(arg)
+        if hasattr(ds, "values"):  # DatasetDict: use train split
+            ds = ds.get("train") or next(iter(ds.values()))
+        for row in ds:
+            yield row.get("text", "")
+    elif kind == "synthetic":
+        rng = np.random.default_rng(seed)
+        words = [f"w{i}" for i in range(2000)]
+        for _ in range(int(arg)):
+            n = int(rng.integers(20, 200))
+            idx = rng.zipf(1.5, size=n).clip(1, len(words)) - 1
+            yield " ".join(words[i] for i in idx)
+    else:
+        raise ValueError(f"unknown source kind {kind!r}")
+
+
+def convert(
+    source: str,
+    out_root: str | Path,
+    num_clients: int = 8,
+    concat_tokens: int = 2048,
+    split: str = "train",
+    tokenizer_spec: str | None = None,
+    val_fraction: float = 0.0,
+    seed: int = 1337,
+) -> dict:
+    """Tokenize + concat documents (EOS-joined) and round-robin them across
+    ``num_clients`` shard dirs, tracking per-client unigram counts."""
+    out_root = Path(out_root)
+    tok = load_tokenizer(tokenizer_spec)
+    eos = getattr(tok, "eos_token_id", None)
+    writers = [
+        TokenShardWriter(out_root / f"client_{i}" / split) for i in range(num_clients)
+    ]
+    freqs = [Counter() for _ in range(num_clients)]
+    totals = [0] * num_clients
+    buf: list[list[int]] = [[] for _ in range(num_clients)]
+    n_docs = 0
+    for doc in iter_documents(source, seed=seed):
+        cid = n_docs % num_clients
+        ids = tok.encode(doc)
+        if eos is not None:
+            ids = ids + [eos]
+        buf[cid].extend(ids)
+        freqs[cid].update(ids)
+        totals[cid] += len(ids)
+        n_docs += 1
+        # flush in concat_tokens blocks (sample granularity of the reader)
+        while len(buf[cid]) >= concat_tokens:
+            writers[cid].write(np.asarray(buf[cid][:concat_tokens]))
+            buf[cid] = buf[cid][concat_tokens:]
+    for cid in range(num_clients):
+        if buf[cid]:
+            writers[cid].write(np.asarray(buf[cid]))
+        writers[cid].close()
+        fdir = out_root / f"client_{cid}"
+        with open(fdir / "1_gram.json", "w") as f:
+            json.dump({str(k): int(v) for k, v in freqs[cid].items()}, f)
+    if tokenizer_spec is not None or True:
+        tok.save_pretrained(out_root / "tokenizer")
+    manifest = {
+        "num_clients": num_clients,
+        "n_documents": n_docs,
+        "tokens_per_client": totals,
+        "concat_tokens": concat_tokens,
+        "split": split,
+        "vocab_size": getattr(tok, "vocab_size", None),
+    }
+    with open(out_root / "manifest.json", "w") as f:
+        json.dump(manifest, f, indent=1)
+    return manifest
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser(description=__doc__)
+    ap.add_argument("--source", required=True)
+    ap.add_argument("--out", required=True)
+    ap.add_argument("--num-clients", type=int, default=8)
+    ap.add_argument("--concat-tokens", type=int, default=2048)
+    ap.add_argument("--split", default="train")
+    ap.add_argument("--tokenizer", default=None)
+    ap.add_argument("--seed", type=int, default=1337)
+    args = ap.parse_args()
+    m = convert(
+        args.source, args.out, args.num_clients, args.concat_tokens,
+        args.split, args.tokenizer, seed=args.seed,
+    )
+    print(json.dumps(m))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,119 @@
+"""Tests: unigram metrics, dataset conversion CLI, partitioner, wire configs."""
+
+import json
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from photon_amd.conf.wire import (
+    EvaluateConfig,
+    FitConfig,
+    get_fit_config,
+)
+from photon_amd.data.convert import ByteTokenizer, convert
+from photon_amd.data.partitioner import partition
+from photon_amd.data.shards import TokenShardDataset
+from photon_amd.metrics import (
+    PureUnigramCrossEntropy,
+    PureUnigramPerplexity,
+    UnigramNormalizedLanguageCrossEntropy,
+    merge_freq_dicts,
+    unigram_tensor_from_freq,
+)
+
+
+# -- unigram metrics --------------------------------------------------------
+def test_unigram_tensor_and_merge():
+    f1 = {"0": 3, "1": 1}
+    f2 = {"1": 1, "2": 2}
+    merged = merge_freq_dicts([f1, f2])
+    assert merged == {"0": 3, "1": 2, "2": 2}
+    p = unigram_tensor_from_freq(merged, vocab_size=4, smoothing=0.0)
+    assert abs(float(p.sum()) - 1.0) < 1e-6
+    assert float(p[0]) == pytest.approx(3 / 7)
+
+
+def test_pure_unigram_ce_matches_closed_form():
+    # uniform unigram over 4 tokens -> CE = log(4) for any labels
+    p = torch.full((4,), 0.25)
+    m = PureUnigramCrossEntropy(p)
+    m.update(torch.tensor([0, 1, 2, 3, 1, 2]))
+    assert m.compute() == pytest.approx(math.log(4), rel=1e-6)
+    pp = PureUnigramPerplexity(p)
+    pp.update(torch.tensor([0, 1]))
+    assert pp.compute() == pytest.approx(4.0, rel=1e-6)
+
+
+def test_unigram_normalized_ce():
+    # model CE == unigram CE -> normalized CE == 0
+    p = torch.full((4,), 0.25)
+    m = UnigramNormalizedLanguageCrossEntropy(p)
+    labels = torch.tensor([0, 1, 2])
+    token_losses = torch.full((3,), math.log(4))
+    m.update(token_losses, labels)
+    assert m.compute() == pytest.approx(0.0, abs=1e-6)
+    # ignore_index masked out
+    m.reset()
+    labels2 = torch.tensor([0, -100, 2])
+    m.update(torch.tensor([1.0, 99.0, 1.0]), labels2)
+    assert m.count == 2
+
+
+# -- conversion CLI ---------------------------------------------------------
+def test_convert_synthetic_and_read(tmp_path):
+    manifest = convert("synthetic:40", tmp_path, num_clients=4,
+                       concat_tokens=128, split="train")
+    assert manifest["n_documents"] == 40
+    for cid in range(4):
+        d = tmp_path / f"client_{cid}" / "train"
+        assert (d / "index.json").exists()
+        freq = json.loads((tmp_path / f"client_{cid}" / "1_gram.json").read_text())
+        assert sum(freq.values()) == manifest["tokens_per_client"][cid]
+        ds = TokenShardDataset(d, seq_len=64)
+        if len(ds) > 0:
+            sample = ds[0]
+            assert sample.shape == (64,)
+            assert int(sample.max()) < ByteTokenizer.vocab_size
+    assert (tmp_path / "tokenizer" / "tokenizer_config.json").exists()
+
+
+def test_partitioner_roundrobin(tmp_path):
+    convert("synthetic:32", tmp_path / "all", num_clients=1,
+            concat_tokens=64, split="train")
+    src = tmp_path / "all" / "client_0" / "train"
+    counts = partition(src, tmp_path / "parts", num_clients=2, block_tokens=64)
+    assert len(counts) == 2
+    assert abs(counts[0] - counts[1]) <= 64
+    ds0 = TokenShardDataset(tmp_path / "parts" / "client_0" / "train", seq_len=64)
+    assert len(ds0) == counts[0] // 64
+
+
+# -- wire configs -----------------------------------------------------------
+def test_fit_config_record_roundtrip():
+    fc = FitConfig(server_round=3, client_ids=[1, 5], local_steps="20ba",
+                   frozen_layers=["transformer.wte.weight"])
+    rec = fc.to_record()
+    assert isinstance(rec["client_ids"], str)
+    fc2 = FitConfig.from_record(rec)
+    assert fc2.client_ids == [1, 5]
+    assert fc2.local_steps == 20
+    assert fc2.frozen_layers == ["transformer.wte.weight"]
+
+
+def test_get_fit_config_from_cfg():
+    cfg = {
+        "fl": {"reset_optimizer": False, "aggregate_momenta": True,
+               "personalized_layers": ["wte"]},
+        "llm_config": {"local_steps": "8ba"},
+    }
+    fc = get_fit_config(cfg, 2, [0, 1], server_steps_cumulative=16)
+    assert fc.local_steps == 8 and not fc.reset_optimizer
+    assert fc.aggregate_momenta and fc.personalized_layers == ["wte"]
+
+
+def test_evaluate_config_defaults():
+    ec = EvaluateConfig(server_round=1, client_ids="[0, 2]")
+    assert ec.client_ids == [0, 2]
+    assert ec.eval_subset_num_batches == -1
