@@ -73,6 +73,10 @@ class MPTAttention(nn.Module):
         super().__init__()
         self.cfg = cfg
         bias = not cfg.no_bias
+        # instance attrs (not cfg reads) so tensor parallelism can shard the
+        # head dimension per rank (photon_amd.parallel.tp).
+        self.n_heads = cfg.n_heads
+        self.d_head = cfg.d_head
         self.Wqkv = nn.Linear(cfg.d_model, 3 * cfg.d_model, bias=bias)
         self.out_proj = nn.Linear(cfg.d_model, cfg.d_model, bias=bias)
         slopes = alibi_slopes(cfg.n_heads, cfg.alibi_bias_max)
@@ -80,17 +84,17 @@ class MPTAttention(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, D = x.shape
-        H, dh = self.cfg.n_heads, self.cfg.d_head
+        H, dh = self.n_heads, self.d_head
         qkv = self.Wqkv(x)
         q, k, v = qkv.chunk(3, dim=-1)
-        # [B, S, D] -> [B, H, S, dh]
+        # [B, S, D_local] -> [B, H, S, dh]
         q = q.view(B, S, H, dh).transpose(1, 2)
         k = k.view(B, S, H, dh).transpose(1, 2)
         v = v.view(B, S, H, dh).transpose(1, 2)
         out = flash_attention(
             q, k, v, self.slopes, causal=True, impl=self.cfg.attn_impl
         )  # [B, H, S, dh]
-        out = out.transpose(1, 2).reshape(B, S, D)
+        out = out.transpose(1, 2).reshape(B, S, H * dh)
         return self.out_proj(out)
 
 
