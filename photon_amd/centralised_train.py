@@ -24,29 +24,16 @@ from .fed.flat import FlatParams
 from .fed.runtime import Comm, init_distributed
 from .history import History
 from .models import build_model
+from .parallel import BucketedGradSync, apply_fsdp, apply_tensor_parallel
 from .train import Trainer
 
 
 def make_grad_sync_hook(comm: Comm):
-    """Bucketed gradient all-reduce (mean) over RCCL after local backward."""
+    """Bucketed gradient all-reduce (mean) over RCCL after local backward
+    (photon_amd.parallel.ddp — FORCED_SYNC semantics)."""
     if not comm.is_distributed:
         return None
-
-    def hook(model: torch.nn.Module) -> None:
-        grads = [p.grad for p in model.parameters() if p.grad is not None]
-        if not grads:
-            return
-        # one flat all-reduce per dtype bucket; ~bucket the lot (model sizes
-        # here are small enough that a single flat is the fastest on xGMI)
-        flat = torch.cat([g.reshape(-1) for g in grads])
-        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
-        flat.div_(comm.world_size)
-        off = 0
-        for g in grads:
-            g.copy_(flat[off : off + g.numel()].view_as(g))
-            off += g.numel()
-
-    return hook
+    return BucketedGradSync(world_size=comm.world_size)
 
 
 def main(argv: list[str] | None = None):
@@ -70,6 +57,15 @@ def main(argv: list[str] | None = None):
     torch.manual_seed(int(llm.get("seed", 17)))
     model = build_model(llm)
 
+    # parallelism config (reference parallelism_config={"fsdp","tp"},
+    # trainer_utils.py:1641-1648): TP shards linears/heads; FSDP wraps
+    # blocks; both degenerate to no-ops at world 1 / falsy config.
+    tp_cfg = llm.get("tp_config") or {}
+    tp_degree = int(tp_cfg.get("degree", 1) or 1)
+    if tp_degree > 1:
+        apply_tensor_parallel(model, rank=rank % tp_degree, world=tp_degree)
+    model = apply_fsdp(model, llm.get("fsdp_config"), device=None)
+
     cent = cfg.get("centralized", {}) or {}
     stream_id = cent.get("stream_id")
     run_uuid = str(cfg.get("run_uuid", "run"))
@@ -91,9 +87,24 @@ def main(argv: list[str] | None = None):
     comm.broadcast_flat(layout.flat, src=0)
     layout.copy_to_model(model)
 
+    # pretrained init: Composer .pt checkpoint, or .npz parameter dump
+    # (reference centralised_train.py:98-117, incl. the WTE-only transplant).
     pretrained = cfg.get("pretrained_model_path")
     if pretrained:
-        trainer.load_checkpoint(pretrained)
+        if str(pretrained).endswith(".npz"):
+            pt_layout = FlatParams(model, filter_key=None, device=device)
+            arrays = pt_layout.load_npz(pretrained)
+            if cent.get("wte_only", False):
+                params = dict(model.named_parameters())
+                for n, a in zip(pt_layout.names, arrays):
+                    if "wte" in n:
+                        with torch.no_grad():
+                            params[n].data.copy_(torch.from_numpy(a).to(params[n].dtype))
+            else:
+                pt_layout.from_ndarrays(arrays)
+                pt_layout.copy_to_model(model)
+        else:
+            trainer.load_checkpoint(pretrained)
 
     history = History(run_dir=save_path / run_uuid if rank == 0 else None,
                       use_wandb=bool(cfg.get("use_wandb", False)))

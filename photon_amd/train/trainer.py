@@ -26,6 +26,8 @@ import torch
 from ..conf.schema import duration_to_batches
 from ..ops.clip import clip_grad_norm_
 from ..ops.optim import build_optimizer
+from .monitors import build_monitors
+from .profiler import maybe_profile
 from .scheduler import build_scheduler
 from .timestamp import Timestamp
 
@@ -74,6 +76,10 @@ class Trainer:
         )
         self.max_duration = duration_to_batches(llm_config.get("max_duration", "1000000ba"))
         self.metrics: dict[str, float] = {}
+        # reference default callback set (speed/lr/memory/runtime/optimizer
+        # monitors, mpt-125m.yaml:98-109)
+        self.monitors = build_monitors(llm_config.get("callbacks"))
+        self.profiler_cfg = llm_config.get("profiler")
 
     # -- precision ----------------------------------------------------------
     def autocast(self):
@@ -125,16 +131,21 @@ class Trainer:
         n_batches = duration_to_batches(duration_batches)
         t0 = time.time()
         losses = []
-        for _ in range(n_batches):
-            if self.timestamp.batch >= self.max_duration:
-                break
-            mbs = [
-                self.train_loader.next_batch() for _ in range(self.grad_accum)
-            ]
-            loss = self.train_batch(mbs)
-            losses.append(loss)
-            if callback is not None:
-                callback(self, loss)
+        with maybe_profile(self.profiler_cfg) as prof:
+            for _ in range(n_batches):
+                if self.timestamp.batch >= self.max_duration:
+                    break
+                mbs = [
+                    self.train_loader.next_batch() for _ in range(self.grad_accum)
+                ]
+                loss = self.train_batch(mbs)
+                losses.append(loss)
+                for mon in self.monitors:
+                    self.metrics.update(mon.batch_end(self, loss))
+                if prof is not None:
+                    prof.step()
+                if callback is not None:
+                    callback(self, loss)
         fit_time = time.time() - t0
         self.metrics.update(
             {

@@ -179,3 +179,24 @@ def test_momenta_export_import():
     pp = o2.param_groups[0]["params"][0]
     assert torch.equal(o2.state[pp]["exp_avg"], m1[0])
     assert o2.state[pp]["step"] == 5
+
+
+def test_monitors_and_profiler(tiny_llm_config, tmp_path):
+    """speed/lr/memory monitors populate metrics; profiler writes a trace."""
+    cfg = dict(tiny_llm_config)
+    cfg["callbacks"] = {
+        "speed_monitor": {"window_size": 4},
+        "lr_monitor": {},
+        "memory_monitor": {},
+        "runtime_estimator": {},
+        "optimizer_monitor": {"interval": 1},
+    }
+    cfg["profiler"] = {"folder": str(tmp_path / "traces"),
+                       "schedule": {"wait": 0, "warmup": 0, "active": 2}}
+    tr = make_trainer(cfg, tmp_path)
+    tr.fit("3ba")
+    keys = tr.metrics.keys()
+    assert "throughput/tokens_per_sec" in keys
+    assert any(k.startswith("lr-") for k in keys)
+    assert "optimizer/l2_norm_grad" in keys
+    assert list((tmp_path / "traces").glob("*.json*")), "chrome trace written"
