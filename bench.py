@@ -37,7 +37,7 @@ def main() -> None:
     ap.add_argument("--model", default="mpt-125m")
     ap.add_argument("--seq-len", type=int, default=2048)
     ap.add_argument("--global-batch", type=int, default=256)
-    ap.add_argument("--microbatch", type=int, default=16)
+    ap.add_argument("--microbatch", type=int, default=32)
     ap.add_argument("--attn", default="flash", choices=["flash", "torch"])
     args = ap.parse_args()
 

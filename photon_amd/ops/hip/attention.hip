@@ -16,7 +16,11 @@ static int attn_lds_bytes(int D, int kind) {
   // kind 1: dq (k + v row images), 2: dkdv (q + do row images + lse/delta).
   // Transposed A-fragments are hardware tr16 reads — no transposed images.
   const int img = 64 * D;  // bytes of one [32][D] bf16 image
-  int imgs = kind == 0 ? 4 * (KBF * D * 2) : (kind == 1 ? 2 * img : 2 * img + 256);
+  // dkdv: double-buffered [QTF][D] q + do images + 2*QTF lse/delta floats
+  // (QTF = 64 at D<=64, 32 at D=128 — mirrors the kernel's constexpr)
+  const int qtf = D <= 64 ? 64 : 32;
+  int imgs = kind == 0 ? 4 * (KBF * D * 2)
+                       : (kind == 1 ? 2 * img : 4 * (qtf * D * 2) + 4 * qtf * 4);
   int bounce = WAVES * img;
   return std::max(imgs, bounce);
 }
