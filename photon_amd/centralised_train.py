@@ -36,6 +36,27 @@ def make_grad_sync_hook(comm: Comm):
     return BucketedGradSync(world_size=comm.world_size)
 
 
+def run_icl_eval(cfg, model, device) -> dict:
+    """Config-gated ICL + gauntlet evaluation (icl_tasks_config /
+    eval_gauntlet_config groups, both `empty` by default)."""
+    icl_cfg = cfg.get("icl_tasks_config") or {}
+    tasks = icl_cfg.get("icl_tasks")
+    if not tasks:
+        return {}
+    from .data.convert import load_tokenizer
+    from .eval import evaluate_icl_tasks, gauntlet_composite
+
+    tok = load_tokenizer(icl_cfg.get("tokenizer_path"))
+    results = evaluate_icl_tasks(
+        model, tasks, tok, device=device,
+        max_seq_len=int(cfg["llm_config"].get("max_seq_len", 2048)),
+        limit_examples=icl_cfg.get("limit_examples"),
+    )
+    gauntlet = (cfg.get("eval_gauntlet_config") or {}).get("eval_gauntlet")
+    results.update(gauntlet_composite(results, gauntlet))
+    return results
+
+
 def main(argv: list[str] | None = None):
     overrides = list(sys.argv[1:] if argv is None else argv)
     save_path = Path(os.environ.get("PHOTON_SAVE_PATH", "."))
@@ -111,6 +132,7 @@ def main(argv: list[str] | None = None):
 
     if cent.get("eval_only", False):
         metrics = trainer.eval()
+        metrics.update(run_icl_eval(cfg, model, device))
         if rank == 0:
             history.add_metrics_centralized(0, metrics)
             print(metrics)
@@ -122,6 +144,7 @@ def main(argv: list[str] | None = None):
     duration = llm.get("max_duration", "100ba")
     metrics = trainer.fit(duration)
     eval_metrics = trainer.eval(int(llm.get("eval_subset_num_batches", -1)))
+    eval_metrics.update(run_icl_eval(cfg, model, device))
     if rank == 0:
         history.add_metrics_centralized(trainer.timestamp.batch, {**metrics, **eval_metrics})
         if cent.get("store_final_model", False):

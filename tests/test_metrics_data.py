@@ -117,3 +117,37 @@ def test_evaluate_config_defaults():
     ec = EvaluateConfig(server_round=1, client_ids="[0, 2]")
     assert ec.client_ids == [0, 2]
     assert ec.eval_subset_num_batches == -1
+
+
+# -- ICL harness ------------------------------------------------------------
+def test_icl_language_modeling_and_mc(tmp_path):
+    from photon_amd.data.convert import ByteTokenizer
+    from photon_amd.eval import evaluate_icl_tasks, gauntlet_composite
+    from photon_amd.models.mpt import MPTCausalLM, MPTConfig
+
+    torch.manual_seed(0)
+    model = MPTCausalLM(
+        MPTConfig(d_model=64, n_heads=2, n_layers=2, max_seq_len=128,
+                  vocab_size=258, attn_impl="torch", loss_impl="torch")
+    )
+    lm = tmp_path / "lm.jsonl"
+    lm.write_text('{"context": "the sky is", "continuation": "blue"}\n'
+                  '{"context": "two plus two is", "continuation": "four"}\n')
+    mc = tmp_path / "mc.jsonl"
+    mc.write_text(json.dumps({"query": "q", "choices": ["a", "b"], "gold": 0}) + "\n")
+    tasks = [
+        {"label": "lm_task", "dataset_uri": str(lm),
+         "icl_task_type": "language_modeling"},
+        {"label": "mc_task", "dataset_uri": str(mc),
+         "icl_task_type": "multiple_choice"},
+    ]
+    res = evaluate_icl_tasks(model, tasks, ByteTokenizer(), max_seq_len=128)
+    assert "metrics/icl/lm_task/accuracy" in res
+    assert 0.0 <= res["metrics/icl/mc_task/accuracy"] <= 1.0
+    gauntlet = {"categories": [
+        {"name": "world_knowledge",
+         "benchmarks": [{"name": "lm_task", "weight": 1.0},
+                        {"name": "mc_task", "weight": 2.0}]},
+    ]}
+    comp = gauntlet_composite(res, gauntlet)
+    assert "metrics/eval_gauntlet/average" in comp
