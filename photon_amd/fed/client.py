@@ -44,6 +44,14 @@ class FedClient:
         self.model = build_model(llm)
         self.trainer: Trainer | None = None
         self.rank = rank
+        from pathlib import Path
+
+        photon = cfg.get("photon", {}) or {}
+        self.save_root = (
+            Path(photon.get("saving_path") or "checkpoints")
+            / str(cfg.get("run_uuid", "run"))
+        )
+        self.save_client_checkpoints = bool(photon.get("checkpoint", False))
         self.client_states: dict[int, ClientState] = {}
         # Per-client persistent trainer state (the reference restores the
         # client's timestamp + dataset_state at every fit,
@@ -104,6 +112,39 @@ class FedClient:
         if cid in self._loader_states and not fl.get("reset_dataset_state", False):
             trainer.train_loader.load_state_dict(self._loader_states[cid])
 
+        # Mid-round resume: if a client checkpoint already holds exactly
+        # steps_done + local_steps batches, load it and SKIP the fit
+        # (reference llm_config_functions.py:642-764, clients/utils.py:217-228).
+        ckpt_dir = self.save_root / f"client_{cid}"
+        expect = trainer.timestamp.batch + steps
+        skip_path = ckpt_dir / f"ep0-ba{expect}-rank{self.rank}.pt"
+        if self.save_client_checkpoints and skip_path.exists():
+            trainer.save_folder = ckpt_dir
+            trainer.load_checkpoint(
+                skip_path, load_ignore_keys=["*scheduler*"]
+            )
+            local_flat = torch.zeros_like(layout.flat)
+            out_views = layout.layer_views_of(local_flat)
+            params = dict(self.model.named_parameters())
+            with torch.no_grad():
+                for n, v in zip(layout.names, out_views):
+                    v.copy_(params[n].detach().to(torch.float32))
+            n_samples = float(steps * int(llm.get("global_train_batch_size", 256)))
+            st = self.client_states.setdefault(cid, ClientState(cid))
+            st.steps_done = trainer.timestamp.batch
+            self._timestamps[cid] = trainer.timestamp.state_dict()
+            momenta_out = bool(fl.get("aggregate_momenta", False))
+            out_payload, pp_metrics = post_process_client_result(
+                layout, local_flat, local_flat, n_samples,
+                trainer=trainer, aggregate_momenta=momenta_out,
+                report_layer_norms=False,
+            )
+            return out_payload, n_samples, {
+                "steps_done": st.steps_done,
+                "client/fit_skipped_from_checkpoint": 1.0,
+                **pp_metrics,
+            }
+
         t0 = time.time()
         global_flat, m1_in, m2_in = manipulate_pre_training(
             payload, layout, fl, cid, local_params=self._personal.get(cid)
@@ -144,6 +185,10 @@ class FedClient:
         self._loader_states[cid] = trainer.train_loader.state_dict()
         if fl.get("personalized_layers"):
             self._personal[cid] = local_flat.clone()
+        if self.save_client_checkpoints:
+            # Composer-format client checkpoint (client_{cid}/ep{e}-ba{b}-
+            # rank{r}.pt) — what the skip-and-load path above reads.
+            trainer.save_checkpoint(ckpt_dir)
 
         out_payload, pp_metrics = post_process_client_result(
             layout, global_flat, local_flat, n_samples,

@@ -235,7 +235,9 @@ class FedServer:
 
     def evaluate_round(self, server_round: int) -> float:
         """Reference evaluate_round (evaluate_utils.py:232): every rank
-        evaluates its client shard; weighted_loss_avg across ranks."""
+        evaluates its client shard; weighted_loss_avg across ranks. With
+        fl.split_eval each client's own stream is also reported per-cid
+        (the reference's split evaluation)."""
         subset = int(self.cfg["llm_config"].get("eval_subset_num_batches", -1))
         if subset <= 0:
             subset = 8
@@ -248,8 +250,15 @@ class FedServer:
         avg = weighted_loss_avg(list(zip(losses, weights)))
         if self.comm.rank == 0:
             self.history.add_loss_distributed(server_round, avg)
+            extra = {}
+            if self.cfg["fl"].get("split_eval", False):
+                extra = {
+                    f"metrics/eval/LanguageCrossEntropy_client_{i}": l
+                    for i, l in enumerate(losses)
+                }
             self.history.add_metrics_distributed(
-                server_round, {"metrics/eval/LanguageCrossEntropy_avg": avg}
+                server_round,
+                {"metrics/eval/LanguageCrossEntropy_avg": avg, **extra},
             )
         return avg
 
@@ -257,7 +266,12 @@ class FedServer:
     def run(self, n_rounds: int | None = None) -> History:
         self.initialize()
         last = self.start_round + (n_rounds or self.n_rounds) - 1
+        refresh = int(self.cfg["photon"].get("refresh_period", 0) or 0)
         for r in range(self.start_round, last + 1):
+            if refresh > 0 and r > self.start_round and (r - 1) % refresh == 0:
+                # prophylactic trainer refresh (reference worker respawn
+                # every photon.refresh_period rounds, client_app.py:175-177)
+                self.client.trainer = None
             self.run_round(r)
             if self.eval_period > 0 and r % self.eval_period == 0:
                 self.evaluate_round(r)

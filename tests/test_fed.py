@@ -162,3 +162,23 @@ def test_two_rank_gloo_equivalence(tiny_cfg, tmp_path):
     assert torch.allclose(results[0], single, atol=1e-6), (
         (results[0] - single).abs().max()
     )
+
+
+def test_client_checkpoint_skip_and_load(tiny_cfg, tmp_path):
+    """Reference mid-round resume: a second fit over the same rounds loads
+    the existing client checkpoint instead of re-training
+    (llm_config_functions.py:642-764)."""
+    import copy
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    m1 = srv.run_round(1)
+    assert "client/fit_skipped_from_checkpoint" not in m1
+    # new server over the same save dir: round 1 must be skipped-from-ckpt
+    srv2 = FedServer(copy.deepcopy(cfg), Comm(0, 1), "cpu")
+    srv2.initialize()
+    m2 = srv2.run_round(1)
+    assert m2.get("client/fit_skipped_from_checkpoint") == 1.0
