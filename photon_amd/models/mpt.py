@@ -24,7 +24,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.attention import alibi_slopes, flash_attention
+from ..ops.attention import alibi_slopes, flash_attention, flash_attention_qkv
 from ..ops.cross_entropy import fused_cross_entropy
 from ..ops.layernorm import FusedLayerNorm
 
@@ -86,15 +86,12 @@ class MPTAttention(nn.Module):
         B, S, D = x.shape
         H, dh = self.n_heads, self.d_head
         qkv = self.Wqkv(x)
-        q, k, v = qkv.chunk(3, dim=-1)
-        # [B, S, D_local] -> [B, H, S, dh]
-        q = q.view(B, S, H, dh).transpose(1, 2)
-        k = k.view(B, S, H, dh).transpose(1, 2)
-        v = v.view(B, S, H, dh).transpose(1, 2)
-        out = flash_attention(
-            q, k, v, self.slopes, causal=True, impl=self.cfg.attn_impl
-        )  # [B, H, S, dh]
-        out = out.transpose(1, 2).reshape(B, S, H * dh)
+        # Packed path: the HIP kernels read [B,S,3,H,dh] strided directly
+        # (no chunk/transpose/contiguous copies); falls back to reshape +
+        # SDPA on CPU / attn_impl=torch.
+        out = flash_attention_qkv(
+            qkv, H, self.slopes, causal=True, impl=self.cfg.attn_impl
+        )  # [B, S, H*dh]
         return self.out_proj(out)
 
 

@@ -44,6 +44,45 @@ def alibi_bias(
     return (-slopes.to(device=device, dtype=dtype).view(1, -1, 1, 1)) * rel
 
 
+class _FlashAttentionQKV(torch.autograd.Function):
+    """Packed-qkv path: takes the fused Wqkv output [B, S, 3*H*dh] and
+    returns [B, S, H*dh] — no transpose/contiguous copies on either side
+    (profiles/r01: the chunk/cat copies were ~3% of step time)."""
+
+    @staticmethod
+    def forward(ctx, qkv, n_heads, slopes, causal):
+        ext = hip_ext()
+        o, lse = ext.attn_fwd_qkv(qkv, n_heads, slopes, causal)
+        ctx.save_for_backward(qkv, slopes, o, lse)
+        ctx.causal = causal
+        ctx.n_heads = n_heads
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, slopes, o, lse = ctx.saved_tensors
+        ext = hip_ext()
+        dqkv = ext.attn_bwd_qkv(
+            do.contiguous(), qkv, ctx.n_heads, slopes, o, lse, ctx.causal
+        )
+        return dqkv, None, None, None
+
+
+def flash_attention_qkv(
+    qkv: torch.Tensor, n_heads: int, slopes: torch.Tensor, causal: bool = True,
+    impl: str = "flash",
+) -> torch.Tensor:
+    """Attention on the packed [B, S, 3*H*dh] Wqkv output; returns
+    [B, S, H*dh]. Falls back to the reshape + SDPA path off-GPU."""
+    B, S, three_hd = qkv.shape
+    dh = three_hd // (3 * n_heads)
+    if impl == "flash" and qkv.dtype == torch.bfloat16 and use_hip(qkv) and dh in (64, 128):
+        return _FlashAttentionQKV.apply(qkv, n_heads, slopes, causal)
+    q, k, v = qkv.view(B, S, 3, n_heads, dh).permute(2, 0, 3, 1, 4).unbind(0)
+    out = flash_attention(q, k, v, slopes, causal=causal, impl=impl)
+    return out.transpose(1, 2).reshape(B, S, n_heads * dh)
+
+
 class _FlashAttentionHIP(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, slopes, causal):

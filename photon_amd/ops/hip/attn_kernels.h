@@ -156,12 +156,17 @@ constexpr int KBF = 64;  // fwd kv-tile keys
 #define ATT_FWD_MINWAVES 2  // min waves/SIMD: caps register alloc at 256
 #endif
 
+// Generalized addressing: element (b, h, s, d) of an input lives at
+// b*bs + h*hs + s*rs + d. Contiguous [B,H,S,D]: (H*S*D, S*D, D). Packed
+// [B,S,3,H,D] qkv (the Wqkv output, no .contiguous() copies): q/k/v
+// pointers pre-offset by their section, strides (S*3*H*D, D, 3*H*D).
 template <int D>
 __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     const __bf16* __restrict__ q, const __bf16* __restrict__ k,
     const __bf16* __restrict__ v, const float* __restrict__ slopes,
     __bf16* __restrict__ out, float* __restrict__ lse_out, int S, int H,
-    int causal) {
+    int causal, long bs_i, long hs_i, long rs_i, long bs_o, long hs_o,
+    long rs_o) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // Per buffer: k row image [KBF][D] + v transposed image [D][KBF].
   constexpr int IMG = KBF * D * 2;  // bytes per image
@@ -179,14 +184,15 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   const float slope = slopes[h];
   const float scale = rsqrtf((float)D);
 
-  const long base = bh * (long)S * D;
+  const long ibase = (bh / H) * bs_i + (long)h * hs_i;
+  const long obase = (bh / H) * bs_o + (long)h * hs_o;
   const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
   const int my_q = q0 + lq;  // this lane's q row
 
   // Q fragments in registers: B-operand, frag kk covers dh [kk*16, kk*16+16)
   bf16x8 qfrag[D / 16];
   {
-    const long qrow = base + (long)min(my_q, S - 1) * D;
+    const long qrow = ibase + (long)min(my_q, S - 1) * rs_i;
 #pragma unroll
     for (int kk = 0; kk < D / 16; ++kk) {
       if (my_q < S) {
@@ -222,8 +228,8 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
       const long grow = (long)t * KBF + row;
 #ifndef ABENCH_NO_LOAD
       if (grow < S) {
-        k_stage[i] = *(const bf16x8*)(k + base + grow * D + col);
-        v_stage[i] = *(const bf16x8*)(v + base + grow * D + col);
+        k_stage[i] = *(const bf16x8*)(k + ibase + grow * rs_i + col);
+        v_stage[i] = *(const bf16x8*)(v + ibase + grow * rs_i + col);
       } else {
         k_stage[i] = bf16x8{};
         v_stage[i] = bf16x8{};
@@ -394,7 +400,7 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     const int row = c / (D / 8);
     const int col = (c % (D / 8)) * 8;
     if (q0 + row < S) {
-      *(bf16x8*)(out + base + (long)(q0 + row) * D + col) =
+      *(bf16x8*)(out + obase + (long)(q0 + row) * rs_o + col) =
           *(const bf16x8*)(o_img + row * D + col);
     }
   }
@@ -410,7 +416,8 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const float* __restrict__ slopes, const float* __restrict__ lse,
     const float* __restrict__ delta, __bf16* __restrict__ dq, int S, int H,
-    int causal) {
+    int causal, long bs_i, long hs_i, long rs_i, long bs_o, long hs_o,
+    long rs_o) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* k_img = (__bf16*)smem;             // [32][D] row image
   __bf16* v_img = (__bf16*)(smem + 64 * D);  // [32][D] row image
@@ -424,18 +431,20 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
   const int h = bh % H;
   const float slope = slopes[h];
   const float scale = rsqrtf((float)D);
-  const long base = bh * (long)S * D;
+  const long ibase = (bh / H) * bs_i + (long)h * hs_i;
+  const long obase = (bh / H) * bs_o + (long)h * hs_o;
   const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
   const int my_q = q0 + lq;
 
   bf16x8 qfrag[D / 16], dofrag[D / 16];
   {
-    const long row = base + (long)min(my_q, S - 1) * D;
+    const long qrow = ibase + (long)min(my_q, S - 1) * rs_i;
+    const long orow = obase + (long)min(my_q, S - 1) * rs_o;
 #pragma unroll
     for (int kk = 0; kk < D / 16; ++kk) {
       if (my_q < S) {
-        qfrag[kk] = *(const bf16x8*)(q + row + kk * 16 + 8 * hi);
-        dofrag[kk] = *(const bf16x8*)(dout + row + kk * 16 + 8 * hi);
+        qfrag[kk] = *(const bf16x8*)(q + qrow + kk * 16 + 8 * hi);
+        dofrag[kk] = *(const bf16x8*)(dout + orow + kk * 16 + 8 * hi);
       } else {
         qfrag[kk] = bf16x8{};
         dofrag[kk] = bf16x8{};
@@ -455,8 +464,8 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
 
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * KB;
-    stage_tile<D>(k + base, kv0, S, D, k_img, nullptr);
-    stage_tile<D>(v + base, kv0, S, D, v_img, nullptr);
+    stage_tile<D>(k + ibase, kv0, S, rs_i, k_img, nullptr);
+    stage_tile<D>(v + ibase, kv0, S, rs_i, v_img, nullptr);
     __syncthreads();
 
     const bool active = !causal || (kv0 <= my_q_max);
@@ -529,7 +538,7 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
     const int row = c / (D / 8);
     const int col = (c % (D / 8)) * 8;
     if (q0 + row < S) {
-      *(bf16x8*)(dq + base + (long)(q0 + row) * D + col) =
+      *(bf16x8*)(dq + ibase + (long)(q0 + row) * rs_i + col) =
           *(const bf16x8*)(o_img + row * D + col);
     }
   }
@@ -549,7 +558,8 @@ void attn_bwd_dkdv_kernel(
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const float* __restrict__ slopes, const float* __restrict__ lse,
     const float* __restrict__ delta, __bf16* __restrict__ dk,
-    __bf16* __restrict__ dv, int S, int H, int causal) {
+    __bf16* __restrict__ dv, int S, int H, int causal, long bs_i, long hs_i,
+    long rs_i, long bs_o, long hs_o, long rs_o) {
   // v2: double-buffered Q/dO tiles, ONE barrier per tile, register-staged
   // loads (same pipeline as attn_fwd_kernel), Q^T/dO^T A-fragments via
   // hardware tr16 reads from the row images. Tile height shrinks at D=128
@@ -572,14 +582,15 @@ void attn_bwd_dkdv_kernel(
   const int h = bh % H;
   const float slope = slopes[h];
   const float scale = rsqrtf((float)D);
-  const long base = bh * (long)S * D;
+  const long ibase = (bh / H) * bs_i + (long)h * hs_i;
+  const long obase = (bh / H) * bs_o + (long)h * hs_o;
   const int k0 = blockIdx.x * (WAVES * KB) + wave * KB;
   const int my_key = k0 + lq;
 
   // K, V rows of this wave's keys as B-operand fragments (like Q in fwd)
   bf16x8 kfrag[D / 16], vfrag[D / 16];
   {
-    const long row = base + (long)min(my_key, S - 1) * D;
+    const long row = ibase + (long)min(my_key, S - 1) * rs_i;
 #pragma unroll
     for (int kk = 0; kk < D / 16; ++kk) {
       if (my_key < S) {
@@ -617,8 +628,8 @@ void attn_bwd_dkdv_kernel(
       const int col = (c % (D / 8)) * 8;
       const long grow = (long)t * QTF + row;
       if (grow < S) {
-        q_stage[i] = *(const bf16x8*)(q + base + grow * D + col);
-        do_stage[i] = *(const bf16x8*)(dout + base + grow * D + col);
+        q_stage[i] = *(const bf16x8*)(q + ibase + grow * rs_i + col);
+        do_stage[i] = *(const bf16x8*)(dout + obase + grow * rs_o + col);
       } else {
         q_stage[i] = bf16x8{};
         do_stage[i] = bf16x8{};
@@ -742,7 +753,7 @@ void attn_bwd_dkdv_kernel(
       const int row = c / (D / 8);
       const int col = (c % (D / 8)) * 8;
       if (k0 + row < S) {
-        *(bf16x8*)(dst + base + (long)(k0 + row) * D + col) =
+        *(bf16x8*)(dst + ibase + (long)(k0 + row) * rs_i + col) =
             *(const bf16x8*)(o_img + row * D + col);
       }
     }

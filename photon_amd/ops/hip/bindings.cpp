@@ -26,6 +26,11 @@ void adopt_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
 torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> gs);
 void multi_tensor_scale_clip(std::vector<torch::Tensor> gs,
                              torch::Tensor total_norm, double max_norm);
+std::vector<torch::Tensor> attn_fwd_qkv(torch::Tensor qkv, long H,
+                                        torch::Tensor slopes, bool causal);
+torch::Tensor attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv, long H,
+                           torch::Tensor slopes, torch::Tensor o,
+                           torch::Tensor lse, bool causal);
 std::vector<torch::Tensor> attn_fwd_launch(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v,
                                            torch::Tensor slopes, bool causal);
@@ -51,6 +56,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale_clip", &multi_tensor_scale_clip);
   m.def("attn_fwd", &attn_fwd_launch, "flash attention forward (ALiBi fused)");
   m.def("attn_bwd", &attn_bwd_launch, "flash attention backward");
+  m.def("attn_fwd_qkv", &attn_fwd_qkv,
+        "flash attention forward on packed [B,S,3HD] qkv");
+  m.def("attn_bwd_qkv", &attn_bwd_qkv,
+        "flash attention backward on packed qkv; returns dqkv");
   m.def("mfma_probe", &mfma_probe);
   m.def("pack_probe", &pack_probe);
 }

@@ -6,60 +6,80 @@
 // The surrounding GEMMs (hipBLASLt) consume that in-place gradient, so the
 // full-precision logits never round-trip to HBM twice.
 //
-// One 256-thread workgroup per row; 8-wide vector loads; online max+sum in
-// a single pass, second pass writes the gradient. V needs no multiple-of-8
-// padding handled here because vocab 50368 = 8 * 6296.
+// v2 (profiles/r01: v1 ran at 1.4 TB/s): branch-free three-pass structure —
+// (1) pure vector max, (2) e = exp(v - max) summed AND stored in place
+// (ONE exp per element instead of v1's two plus a divergent online-max
+// branch), (3) scale by 1/sum. A row is 100 KB bf16: pass 2/3 re-reads hit
+// the XCD L2, so HBM sees one read + one write per element. All accesses
+// are 8-element vectors (v1 stored scalar 2-byte elements).
 
 #include "host_common.h"
 
 namespace photon_hip {
 
 template <typename T, int BLOCK>
-__global__ void ce_fwd_bwd_kernel(T* __restrict__ logits,
-                                  const long* __restrict__ targets,
-                                  float* __restrict__ losses, long V) {
+__global__ __launch_bounds__(BLOCK) void ce_fwd_bwd_kernel(
+    T* __restrict__ logits, const long* __restrict__ targets,
+    float* __restrict__ losses, long V) {
   __shared__ float scratch[BLOCK / WAVE];
   const long row = blockIdx.x;
   T* lr = logits + row * V;
   const long tgt = targets[row];
+  const float logit_tgt = load_f32<T>(lr, tgt);
 
-  // pass 1: online max & sum(exp(x - max))
-  float m = -INFINITY, s = 0.f;
-  for (long i = threadIdx.x * 8; i < V; i += BLOCK * 8) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      if (i + j < V) {
-        float v = load_f32<T>(lr, i + j);
-        if (v > m) {
-          s *= __expf(m - v);
-          m = v;
-        }
-        s += __expf(v - m);
-      }
-    }
+  // pass 1: max (vector loads, no exp, no branches)
+  float m = -1e30f;
+  for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
+    floatx4 a = load4<T>(lr + i);
+    floatx4 b = load4<T>(lr + i + 4);
+    m = fmaxf(m, fmaxf(fmaxf(a.x, a.y), fmaxf(a.z, a.w)));
+    m = fmaxf(m, fmaxf(fmaxf(b.x, b.y), fmaxf(b.z, b.w)));
   }
-  // combine across threads: global max then rescaled sums
-  float gm = block_reduce_max(m, scratch);
-  s *= __expf(m - gm);
-  float gs = block_reduce_sum(s, scratch);
-  const float lse = gm + __logf(gs);
+  for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK)
+    m = fmaxf(m, load_f32<T>(lr, i));
+  const float gm = block_reduce_max(m, scratch);
+
+  // pass 2: e = exp(v - gm), accumulate sum, store e in place (L2-resident)
+  float s = 0.f;
+  for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
+    floatx4 a = load4<T>(lr + i);
+    floatx4 b = load4<T>(lr + i + 4);
+    floatx4 ea, eb;
+    ea.x = __expf(a.x - gm); ea.y = __expf(a.y - gm);
+    ea.z = __expf(a.z - gm); ea.w = __expf(a.w - gm);
+    eb.x = __expf(b.x - gm); eb.y = __expf(b.y - gm);
+    eb.z = __expf(b.z - gm); eb.w = __expf(b.w - gm);
+    s += ea.x + ea.y + ea.z + ea.w + eb.x + eb.y + eb.z + eb.w;
+    store4<T>(lr + i, ea);
+    store4<T>(lr + i + 4, eb);
+  }
+  for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK) {
+    float e = __expf(load_f32<T>(lr, i) - gm);
+    s += e;
+    store_f32<T>(lr, i, e);
+  }
+  const float gs = block_reduce_sum(s, scratch);
+  const float inv = 1.f / gs;
   if (threadIdx.x == 0) {
-    losses[row] = lse - load_f32<T>(lr, tgt);
+    losses[row] = gm + __logf(gs) - logit_tgt;
   }
   __syncthreads();
 
-  // pass 2: write gradient (softmax - onehot)
-  const float inv = 1.f / gs;
-  for (long i = threadIdx.x * 8; i < V; i += BLOCK * 8) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      if (i + j < V) {
-        float v = load_f32<T>(lr, i + j);
-        float p = __expf(v - gm) * inv;
-        if (i + j == tgt) p -= 1.f;
-        store_f32<T>(lr, i + j, p);
-      }
-    }
+  // pass 3: p = e * inv (- onehot); vector stores
+  for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
+    floatx4 a = load4<T>(lr + i);
+    floatx4 b = load4<T>(lr + i + 4);
+    a.x *= inv; a.y *= inv; a.z *= inv; a.w *= inv;
+    b.x *= inv; b.y *= inv; b.z *= inv; b.w *= inv;
+    store4<T>(lr + i, a);
+    store4<T>(lr + i + 4, b);
+  }
+  for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK) {
+    store_f32<T>(lr, i, load_f32<T>(lr, i) * inv);
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    store_f32<T>(lr, tgt, load_f32<T>(lr, tgt) - 1.f);
   }
 }
 

@@ -20,43 +20,6 @@
 
 namespace photon_hip {
 
-// ---------------------------------------------------------------------------
-// Vector access helpers: 4 elements per lane per chunk.
-// bf16: 8-byte load (shortx4); f32: 16-byte load (floatx4).
-// ---------------------------------------------------------------------------
-template <typename T>
-DEV_INLINE floatx4 load4(const T* p);
-template <>
-DEV_INLINE floatx4 load4<unsigned short>(const unsigned short* p) {
-  shortx4 r = *reinterpret_cast<const shortx4*>(p);
-  floatx4 f;
-  f.x = bf16_to_f32((unsigned short)r.x);
-  f.y = bf16_to_f32((unsigned short)r.y);
-  f.z = bf16_to_f32((unsigned short)r.z);
-  f.w = bf16_to_f32((unsigned short)r.w);
-  return f;
-}
-template <>
-DEV_INLINE floatx4 load4<float>(const float* p) {
-  return *reinterpret_cast<const floatx4*>(p);
-}
-
-template <typename T>
-DEV_INLINE void store4(T* p, floatx4 v);
-template <>
-DEV_INLINE void store4<unsigned short>(unsigned short* p, floatx4 v) {
-  shortx4 r;
-  r.x = (short)f32_to_bf16(v.x);
-  r.y = (short)f32_to_bf16(v.y);
-  r.z = (short)f32_to_bf16(v.z);
-  r.w = (short)f32_to_bf16(v.w);
-  *reinterpret_cast<shortx4*>(p) = r;
-}
-template <>
-DEV_INLINE void store4<float>(float* p, floatx4 v) {
-  *reinterpret_cast<floatx4*>(p) = v;
-}
-
 #define COLS_PER_WAVE 256  // 64 lanes * 4 elements
 
 // ---------------------------------------------------------------------------

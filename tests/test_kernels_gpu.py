@@ -297,3 +297,32 @@ def test_model_train_step_flash_vs_torch(dev, ext):
         ).norm()
         assert torch.isfinite(grad_norm)
     assert abs(losses["flash"] - losses["torch"]) < 0.05, losses
+
+
+@pytest.mark.gpu
+def test_attn_qkv_packed_matches_separate():
+    """Packed [B,S,3HD] path must match the separate-tensor path (fwd+bwd)."""
+    from photon_amd.ops.attention import alibi_slopes, flash_attention, flash_attention_qkv
+
+    torch.manual_seed(11)
+    B, H, S, D = 2, 4, 256, 64
+    qkv = torch.randn(B, S, 3 * H * D, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    slopes = alibi_slopes(H).to("cuda")
+    out_p = flash_attention_qkv(qkv, H, slopes, causal=True)
+    g = torch.randn_like(out_p)
+    out_p.backward(g)
+    dqkv_p = qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q, k, v = qkv2.view(B, S, 3, H, D).permute(2, 0, 3, 1, 4).unbind(0)
+    out_s = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                            slopes, causal=True)
+    out_s = out_s.transpose(1, 2).reshape(B, S, H * D)
+    out_s.backward(g)
+    assert torch.allclose(out_p.float(), out_s.float(), atol=1e-3), (
+        (out_p.float() - out_s.float()).abs().max()
+    )
+    assert torch.allclose(dqkv_p.float(), qkv2.grad.float(), atol=1e-3), (
+        (dqkv_p.float() - qkv2.grad.float()).abs().max()
+    )
