@@ -30,31 +30,37 @@ class _FusedLinearCrossEntropy(torch.autograd.Function):
         N, D = hidden.shape
         V = weight.shape[0]
         device = hidden.device
-        total = torch.zeros((), device=device, dtype=torch.float32)
-        # Pre-allocate grad buffers; filled chunk by chunk (recompute-free
-        # backward: we already know dlogits up to the 1/N * upstream scale).
-        dh = torch.empty_like(hidden)
-        dw = torch.zeros_like(weight, dtype=torch.float32)
-        for s in range(0, N, _CHUNK):
-            e = min(s + _CHUNK, N)
-            h = hidden[s:e]
-            t = targets[s:e]
-            logits = h @ weight.t()  # [c, V] in h.dtype (bf16 on GPU)
-            if use_kernel:
-                ext = hip_ext()
-                # Computes per-row loss (fp32) and overwrites `logits` with
-                # (softmax - onehot) in logits.dtype.
-                losses = ext.ce_fwd_bwd_inplace(logits, t)
-                dl = logits
-            else:
+        if use_kernel:
+            # MI355X path: 288 GB HBM holds the full [N, V] bf16 logits
+            # (6.6 GB at N=65k, V=50368), so run ONE logits GEMM, one CE
+            # kernel sweep (in-place gradient), and ONE GEMM per input grad.
+            # The previous chunked variant paid 2 extra fp32 [V, D]
+            # convert+add passes PER CHUNK for the dw accumulation
+            # (~3.5% of step time, profiles/r01 prof6).
+            ext = hip_ext()
+            logits = hidden @ weight.t()  # [N, V] bf16
+            losses = ext.ce_fwd_bwd_inplace(logits, targets)
+            total = losses.sum()
+            dl = logits  # overwritten with (softmax - onehot)
+            dh = dl @ weight
+            dw = (dl.t() @ hidden).float()
+        else:
+            total = torch.zeros((), device=device, dtype=torch.float32)
+            dh = torch.empty_like(hidden)
+            dw = torch.zeros_like(weight, dtype=torch.float32)
+            for s in range(0, N, _CHUNK):
+                e = min(s + _CHUNK, N)
+                h = hidden[s:e]
+                t = targets[s:e]
+                logits = h @ weight.t()
                 lf = logits.float()
                 losses = F.cross_entropy(lf, t, reduction="none")
                 dl = torch.softmax(lf, dim=-1)
                 dl[torch.arange(e - s, device=device), t] -= 1.0
                 dl = dl.to(logits.dtype)
-            total = total + losses.sum()
-            dh[s:e] = dl @ weight
-            dw += (dl.t() @ h).float()
+                total = total + losses.sum()
+                dh[s:e] = dl @ weight
+                dw += (dl.t() @ h).float()
         ctx.save_for_backward(dh, dw)
         ctx.n_rows = N
         ctx.w_dtype = weight.dtype
