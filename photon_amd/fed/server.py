@@ -40,6 +40,7 @@ from .noise_scale import FedSimpleNoiseScale
 from .runtime import Comm, assign_clients_to_ranks, sample_clients
 from .params_ops import join_payload, split_payload
 from .server_ckpt import (
+    copy_old_checkpoints_to_new_run,
     interpret_resume_round,
     obtain_sorted_rounds,
     resume_from_round,
@@ -113,8 +114,20 @@ class FedServer:
 
     # -- initialization / resume -------------------------------------------
     def initialize(self) -> None:
-        """initialize_round / resume_from_round (init_utils.py:128-287)."""
+        """initialize_round / resume_from_round / restore
+        (init_utils.py:128-287, server_app.py:159-219)."""
         resumed = None
+        photon_cfg = self.cfg["photon"]
+        # cross-run restore: copy an old run's latest complete server round
+        # (+ client checkpoints) under this run's prefix before resuming
+        restore_uuid = photon_cfg.get("restore_run_uuid")
+        if restore_uuid and self.comm.rank == 0:
+            copy_old_checkpoints_to_new_run(
+                self.saving_path, str(restore_uuid), self.run_uuid,
+                self.strategy.state_keys,
+            )
+        if restore_uuid:
+            self.comm.barrier()
         resume_round = self.cfg["photon"].get("resume_round", -1)
         rounds = obtain_sorted_rounds(
             self.saving_path, self.run_uuid, self.strategy.state_keys
@@ -131,7 +144,17 @@ class FedServer:
         else:
             # fresh init: rank 0's model init is the global model; broadcast
             # the flat buffer once (the only full-parameter broadcast).
-            self.layout.copy_from_model(self.client.model)
+            # restore_cent_run_uuid bootstraps from a centralized run's
+            # final_parameters.npz (init_utils.py:43-125).
+            cent_uuid = photon_cfg.get("restore_cent_run_uuid")
+            if cent_uuid:
+                npz = (
+                    Path(self.saving_path) / str(cent_uuid)
+                    / "final_parameters.npz"
+                )
+                self.layout.from_ndarrays(self.layout.load_npz(npz))
+            else:
+                self.layout.copy_from_model(self.client.model)
             self.comm.broadcast_flat(self.layout.flat, src=0)
             self.strategy.initialize(self.layout.flat)
         # all ranks start from identical global params

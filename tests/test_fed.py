@@ -182,3 +182,28 @@ def test_client_checkpoint_skip_and_load(tiny_cfg, tmp_path):
     srv2.initialize()
     m2 = srv2.run_round(1)
     assert m2.get("client/fit_skipped_from_checkpoint") == 1.0
+
+
+def test_restore_run_uuid(tiny_cfg, tmp_path):
+    """Cross-run restore: a new run_uuid picks up the old run's latest
+    server round (s3_utils.py:275-345,1478-1608 semantics)."""
+    import copy
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    cfg["run_uuid"] = "old_run"
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.run(2)
+    old_params = srv.strategy.params.clone()
+
+    cfg2 = copy.deepcopy(tiny_cfg)
+    cfg2["photon"]["checkpoint"] = False
+    cfg2["photon"]["saving_path"] = str(tmp_path)
+    cfg2["photon"]["restore_run_uuid"] = "old_run"
+    cfg2["photon"]["resume_round"] = -1
+    cfg2["run_uuid"] = "new_run"
+    srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
+    srv2.initialize()
+    assert srv2.start_round == 3
+    assert torch.allclose(srv2.strategy.params, old_params, atol=1e-6)
