@@ -19,8 +19,10 @@ static int attn_lds_bytes(int D, int kind) {
   // dkdv: double-buffered [QTF][D] q + do images + 2*QTF lse/delta floats
   // (QTF = 64 at D<=64, 32 at D=128 — mirrors the kernel's constexpr)
   const int qtf = D <= 64 ? 64 : 32;
+  // dq: double-buffered [64][D] k + v row images
   int imgs = kind == 0 ? 4 * (KBF * D * 2)
-                       : (kind == 1 ? 2 * img : 4 * (qtf * D * 2) + 4 * qtf * 4);
+                       : (kind == 1 ? 4 * (64 * D * 2)
+                                    : 4 * (qtf * D * 2) + 4 * qtf * 4);
   int bounce = WAVES * img;
   return std::max(imgs, bounce);
 }

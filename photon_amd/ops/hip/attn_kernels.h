@@ -411,17 +411,23 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
 // dQ^T[dh][q] = sum_key K^T[dh][key] * dS[key][q] * scale
 // ---------------------------------------------------------------------------
 template <int D>
-__global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
+// D=128 needs >256 VGPRs with the staged pipeline; capping at 2 waves/SIMD
+// spills — run 1 wave/SIMD there instead (cf. dkdv).
+__global__ __launch_bounds__(ATT_BLOCK, (D <= 64) ? 2 : 1)
+void attn_bwd_dq_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
     const float* __restrict__ slopes, const float* __restrict__ lse,
     const float* __restrict__ delta, __bf16* __restrict__ dq, int S, int H,
     int causal, long bs_i, long hs_i, long rs_i, long bs_o, long hs_o,
     long rs_o) {
+  // v2: 64-key double-buffered K/V tiles, ONE barrier per tile, register
+  // staging (the fwd pipeline); K^T A-fragments via hardware tr16 reads.
+  constexpr int KBQ = 64;           // kv keys per staged tile (2 subtiles)
+  constexpr int IMG2 = KBQ * D * 2;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* k_img = (__bf16*)smem;             // [32][D] row image
-  __bf16* v_img = (__bf16*)(smem + 64 * D);  // [32][D] row image
-  // K^T A-fragments come from k_img via lds_tr16 (no transposed image).
+  auto k_img = [&](int b) { return (__bf16*)(smem + b * 2 * IMG2); };
+  auto v_img = [&](int b) { return (__bf16*)(smem + b * 2 * IMG2 + IMG2); };
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -458,41 +464,78 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int db = 0; db < D / 32; ++db) dq_acc[db] = f32x16{};
 
-  const int q_max_block = min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
-  const int n_tiles = causal ? (q_max_block / KB + 1) : ((S + KB - 1) / KB);
+  const int q_max_block =
+      min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
+  const int n_tiles = causal ? (q_max_block / KBQ + 1) : ((S + KBQ - 1) / KBQ);
   const int my_q_max = min(q0 + QB - 1, S - 1);
 
+  constexpr int NCHUNK = KBQ * D / 8 / ATT_BLOCK;
+  bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
+  auto stage_load = [&](int t) {
+#pragma unroll
+    for (int i = 0; i < NCHUNK; ++i) {
+      const int c = i * ATT_BLOCK + threadIdx.x;
+      const int row = c / (D / 8);
+      const int col = (c % (D / 8)) * 8;
+      const long grow = (long)t * KBQ + row;
+      if (grow < S) {
+        k_stage[i] = *(const bf16x8*)(k + ibase + grow * rs_i + col);
+        v_stage[i] = *(const bf16x8*)(v + ibase + grow * rs_i + col);
+      } else {
+        k_stage[i] = bf16x8{};
+        v_stage[i] = bf16x8{};
+      }
+    }
+  };
+  auto stage_write = [&](int b) {
+#pragma unroll
+    for (int i = 0; i < NCHUNK; ++i) {
+      const int c = i * ATT_BLOCK + threadIdx.x;
+      const int row = c / (D / 8);
+      const int col = (c % (D / 8)) * 8;
+      lds_store16(k_img(b), row, D * 2, col * 2, k_stage[i]);
+      lds_store16(v_img(b), row, D * 2, col * 2, v_stage[i]);
+    }
+  };
+
+  stage_load(0);
+  stage_write(0);
+
   for (int t = 0; t < n_tiles; ++t) {
-    const int kv0 = t * KB;
-    stage_tile<D>(k + ibase, kv0, S, rs_i, k_img, nullptr);
-    stage_tile<D>(v + ibase, kv0, S, rs_i, v_img, nullptr);
+    const int buf = t & 1;
+    const int kv0 = t * KBQ;
+    if (t + 1 < n_tiles) stage_load(t + 1);
     __syncthreads();
 
-    const bool active = !causal || (kv0 <= my_q_max);
-    if (active) {
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kv0s = kv0 + sub * 32;
+      if (causal && kv0s > my_q_max) break;
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
 #pragma unroll
       for (int kk = 0; kk < D / 16; ++kk) {
-        bf16x8 ka = lds_frag(k_img, lq, D * 2, kk * 32 + hi * 16);
-        bf16x8 va = lds_frag(v_img, lq, D * 2, kk * 32 + hi * 16);
+        bf16x8 ka =
+            lds_frag(k_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
+        bf16x8 va =
+            lds_frag(v_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[kk], s_acc,
                                                         0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[kk],
                                                          dp_acc, 0, 0, 0);
       }
-      // p = exp(sv - lse); masked keys land at exp(-huge) == 0 without a
-      // per-element compare (finite sentinel; cf. fwd).
+      // p = exp(sv - lse); masked keys -> exp(-huge) == 0 (no compares on
+      // non-diagonal tiles; need_mask is BLOCK-uniform).
       const bool need_mask =
-          (kv0 + KB > S) ||
-          (causal && (kv0 + KB - 1 > (int)blockIdx.x * (WAVES * QB)));
-      const float abase = slope * (float)(kv0 - my_q);
+          (kv0s + 32 > S) ||
+          (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
+      const float abase = slope * (float)(kv0s - my_q);
       float ds[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
         float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
         if (need_mask) {
-          const int key = kv0 + pat;
+          const int key = kv0s + pat;
           const bool masked = (key >= S) || (causal && key > my_q);
           sv = masked ? -1e30f : sv;
         }
@@ -504,7 +547,7 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int s16 = 0; s16 < 2; ++s16) {
         bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
-        const int koff = s16 * 16 + 8 * hi + (tj >> 2);
+        const int koff = sub * 32 + s16 * 16 + 8 * hi + (tj >> 2);
 #pragma unroll
         for (int db = 0; db < D / 32; ++db) {
           const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
@@ -513,15 +556,16 @@ __global__ __launch_bounds__(ATT_BLOCK, 2) void attn_bwd_dq_kernel(
           for (int rd = 0; rd < 2; ++rd) {
             const int key = koff + 4 * rd;
             a.h[rd] = lds_tr16(
-                k_img, swz((unsigned)(key * (D * 2) + dhc * 2), key));
+                k_img(buf), swz((unsigned)(key * (D * 2) + dhc * 2), key));
           }
           dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a.v8, dsfrag, dq_acc[db], 0, 0, 0);
         }
       }
     }
-    __syncthreads();
+    if (t + 1 < n_tiles) stage_write(1 - buf);
   }
+  __syncthreads();  // protect epilogue smem reuse
 
   // epilogue (tr-read rows are natural order; see fwd epilogue note)
   __bf16* o_img = (__bf16*)smem + wave * 32 * D;
