@@ -19,4 +19,20 @@ The Hydra config surface (photon/conf/*) and checkpoint formats
 are kept compatible so reference runs can be restored.
 """
 
+# -- GEMM autotuning ---------------------------------------------------------
+# hipBLASLt/rocBLAS solution selection via PyTorch TunableOp, READ-ONLY from
+# the pre-tuned MI355X table (photon_amd/tuned/tunableop_mi355x.csv, +3-5%
+# end-to-end measured). Opt out with PHOTON_NO_TUNABLEOP=1; re-tune with
+# PYTORCH_TUNABLEOP_TUNING=1.
+import os as _os
+
+if _os.environ.get("PHOTON_NO_TUNABLEOP", "0") != "1":
+    _tuned = _os.path.join(_os.path.dirname(__file__), "tuned",
+                           "tunableop_mi355x.csv")
+    if _os.path.exists(_tuned):
+        _os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        _os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+        _os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _tuned)
+
+
 __version__ = "0.1.0"
