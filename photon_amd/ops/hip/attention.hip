@@ -101,6 +101,28 @@ std::vector<torch::Tensor> attn_fwd_qkv(torch::Tensor qkv, long H,
                     causal, B, (int)H, S, D, in, out_s, o, qkv);
 }
 
+static torch::Tensor delta_launch(torch::Tensor dout, torch::Tensor o,
+                                  int B, int H, int S, int D, Strides st) {
+  auto delta = torch::empty({(long)B, (long)H, (long)S},
+                            dout.options().dtype(at::kFloat));
+  const long n_rows = (long)B * S * H;
+  const int rpw = 64 / (D / 4);
+  long want = (n_rows + 4L * rpw - 1) / (4L * rpw);
+  int G = (int)std::min<long>(std::max<long>(want, 64), 2048);
+  if (D == 64) {
+    hipLaunchKernelGGL((attn_delta_kernel<64>), dim3(G), dim3(256), 0,
+                       cur_stream(), (const __bf16*)dout.data_ptr(),
+                       (const __bf16*)o.data_ptr(), delta.data_ptr<float>(),
+                       n_rows, S, H, st.bs, st.hs, st.rs);
+  } else {
+    hipLaunchKernelGGL((attn_delta_kernel<128>), dim3(G), dim3(256), 0,
+                       cur_stream(), (const __bf16*)dout.data_ptr(),
+                       (const __bf16*)o.data_ptr(), delta.data_ptr<float>(),
+                       n_rows, S, H, st.bs, st.hs, st.rs);
+  }
+  return delta;
+}
+
 static void bwd_common(const __bf16* dout, const __bf16* q, const __bf16* k,
                        const __bf16* v, torch::Tensor slopes_f,
                        torch::Tensor lse, torch::Tensor delta, __bf16* dq,
@@ -136,8 +158,8 @@ std::vector<torch::Tensor> attn_bwd_launch(torch::Tensor dout, torch::Tensor q,
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
   auto slopes_f = slopes.to(q.device(), at::kFloat).contiguous();
-  auto delta = (dout.to(at::kFloat) * o.to(at::kFloat)).sum(-1).contiguous();
   auto st = bhsd_strides(H, S, D);
+  auto delta = delta_launch(dout.contiguous(), o, B, H, S, D, st);
   bwd_common((const __bf16*)dout.data_ptr(), (const __bf16*)q.data_ptr(),
              (const __bf16*)k.data_ptr(), (const __bf16*)v.data_ptr(),
              slopes_f, lse, delta, (__bf16*)dq.data_ptr(),
@@ -155,14 +177,10 @@ torch::Tensor attn_bwd_qkv(torch::Tensor dout, torch::Tensor qkv, long H,
   const int D = (int)(qkv.size(2) / (3 * H));
   auto dqkv = torch::empty_like(qkv);
   auto slopes_f = slopes.to(qkv.device(), at::kFloat).contiguous();
-  // delta = rowsum(dO * O) per (b,h,s): [B,S,H*D] -> [B,H,S]
-  auto delta = (dout.to(at::kFloat) * o.to(at::kFloat))
-                   .view({(long)B, (long)S, H, (long)D})
-                   .sum(-1)
-                   .transpose(1, 2)
-                   .contiguous();
   auto in = qkv_strides((int)H, S, D);
   auto out_s = bshd_strides((int)H, S, D);
+  // delta = rowsum(dO * O) per (b,h,s) straight off the packed layout
+  auto delta = delta_launch(dout, o, B, (int)H, S, D, out_s);
   const __bf16* base = (const __bf16*)qkv.data_ptr();
   __bf16* dbase = (__bf16*)dqkv.data_ptr();
   bwd_common((const __bf16*)dout.data_ptr(), base, base + (long)H * D,

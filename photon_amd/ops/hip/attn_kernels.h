@@ -806,4 +806,46 @@ void attn_bwd_dkdv_kernel(
 }
 
 
+// ---------------------------------------------------------------------------
+// delta[b,h,s] = sum_d dO[b,s,h,d] * O[b,s,h,d]  (fp32 out) — replaces the
+// torch chain (two fp32 casts + a reduce kernel) for the FlashAttention-2
+// delta term. 256/D rows per wave: each row is covered by a D/4-lane
+// subgroup loading 4 bf16 per lane; subgroup reduction via shfl_xor.
+// Strides (bs, hs, rs) address both tensors (packed [B,S,H,D] or [B,H,S,D]).
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ __launch_bounds__(256) void attn_delta_kernel(
+    const __bf16* __restrict__ dout, const __bf16* __restrict__ o,
+    float* __restrict__ delta, long n_rows /* B*S*H */, int S, int H,
+    long bs, long hs, long rs) {
+  constexpr int LPR = D / 4;          // lanes per row (16 or 32)
+  constexpr int RPW = 64 / LPR;       // rows per wave (4 or 2)
+  const int lane = threadIdx.x & 63;
+  const int sub = lane / LPR;         // row slot within the wave
+  const int sl = lane % LPR;          // lane within the row subgroup
+  const long wave_id = ((long)blockIdx.x * blockDim.x + threadIdx.x) / 64;
+  const long n_waves = ((long)gridDim.x * blockDim.x) / 64;
+
+  for (long r0 = wave_id * RPW; r0 < n_rows; r0 += n_waves * RPW) {
+    const long idx = r0 + sub;  // this subgroup's (b, s, h) row
+    float acc = 0.f;
+    long out_pos = 0;
+    if (idx < n_rows) {
+      const long b = idx / ((long)S * H);
+      const long rem = idx % ((long)S * H);
+      const long s_ = rem / H;
+      const long h_ = rem % H;
+      const long off = b * bs + h_ * hs + s_ * rs + sl * 4;
+      floatx4 a = load4<unsigned short>((const unsigned short*)(dout + off));
+      floatx4 bb = load4<unsigned short>((const unsigned short*)(o + off));
+      acc = a.x * bb.x + a.y * bb.y + a.z * bb.z + a.w * bb.w;
+      out_pos = (b * H + h_) * (long)S + s_;
+    }
+#pragma unroll
+    for (int offx = LPR / 2; offx > 0; offx >>= 1)
+      acc += __shfl_xor(acc, offx, 64);
+    if (sl == 0 && idx < n_rows) delta[out_pos] = acc;
+  }
+}
+
 }  // namespace photon_hip
