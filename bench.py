@@ -50,6 +50,9 @@ def main() -> None:
     from photon_amd.models import build_model
     from photon_amd.train import Trainer
 
+    from photon_amd.fed.rccl_tuning import apply_rccl_env
+
+    apply_rccl_env(cfg=None)
     rank, world = init_distributed()
     if world == 1 and args.gpus > 1:
         raise SystemExit("multi-GPU bench must be launched via torchrun")

@@ -38,6 +38,9 @@ def main(argv: list[str] | None = None):
     else:
         cfg = validate(compose(config_yaml_dir(), "base", overrides))
 
+    from .fed.rccl_tuning import apply_rccl_env
+
+    apply_rccl_env(cfg)
     rank, world = init_distributed()
     comm = Comm(rank, world)
     device = (
