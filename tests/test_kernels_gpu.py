@@ -326,3 +326,36 @@ def test_attn_qkv_packed_matches_separate():
     assert torch.allclose(dqkv_p.float(), qkv2.grad.float(), atol=1e-3), (
         (dqkv_p.float() - qkv2.grad.float()).abs().max()
     )
+
+
+@pytest.mark.gpu
+def test_ce_kernel_odd_vocab():
+    """CE kernel remainder path: vocab not a multiple of 8."""
+    from photon_amd.ops import hip_ext
+
+    torch.manual_seed(3)
+    x = torch.randn(64, 1003, device="cuda", dtype=torch.bfloat16)
+    t = torch.randint(0, 1003, (64,), device="cuda")
+    ref = torch.nn.functional.cross_entropy(
+        x.float(), t, reduction="none"
+    )
+    losses = hip_ext().ce_fwd_bwd_inplace(x.clone(), t)
+    assert torch.allclose(losses, ref, atol=2e-2), (losses - ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_layernorm_d2560():
+    """LN kernel at MPT-3B's d_model (2560 = 10 chunks/wave)."""
+    from photon_amd.ops.layernorm import FusedLayerNorm
+
+    torch.manual_seed(4)
+    ln = FusedLayerNorm(2560).to("cuda")
+    x = torch.randn(128, 2560, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ln(x)
+    ref = torch.nn.functional.layer_norm(
+        x.float(), (2560,), ln.weight.float(), ln.bias.float(), ln.eps
+    )
+    assert (y.float() - ref).abs().max() < 2e-2
+    y.sum().backward()
+    assert x.grad is not None and ln.weight.grad is not None
