@@ -37,14 +37,17 @@ void bench(int B, int H, int S, int iters) {
   dim3 grid((S + WAVES * QB - 1) / (WAVES * QB), (size_t)B * H);
   const int lds = 4 * KBF * D * 2 > WAVES * 64 * D ? 4 * KBF * D * 2 : WAVES * 64 * D;
   hipEvent_t e0, e1; hipEventCreate(&e0); hipEventCreate(&e1);
+  const long bs0 = (long)H * S * D, hs0 = (long)S * D, rs0 = D;
   for (int i = 0; i < 3; ++i)
     hipLaunchKernelGGL((attn_fwd_kernel<D>), grid, dim3(ATT_BLOCK), lds, 0,
-                       q, k, v, slopes, o, lse, S, H, 1);
+                       q, k, v, slopes, o, lse, S, H, 1, bs0, hs0, rs0,
+                       bs0, hs0, rs0);
   CHECK(hipDeviceSynchronize());
   hipEventRecord(e0);
   for (int i = 0; i < iters; ++i)
     hipLaunchKernelGGL((attn_fwd_kernel<D>), grid, dim3(ATT_BLOCK), lds, 0,
-                       q, k, v, slopes, o, lse, S, H, 1);
+                       q, k, v, slopes, o, lse, S, H, 1, bs0, hs0, rs0, bs0,
+                       hs0, rs0);
   hipEventRecord(e1);
   CHECK(hipDeviceSynchronize());
   float ms; hipEventElapsedTime(&ms, e0, e1); ms /= iters;
@@ -76,8 +79,10 @@ void check(int B, int H, int S) {
   hipMemcpy(slopes, hs.data(), H * 4, hipMemcpyHostToDevice);
   dim3 grid((S + WAVES * QB - 1) / (WAVES * QB), (size_t)B * H);
   const int lds = 4 * KBF * D * 2 > WAVES * 64 * D ? 4 * KBF * D * 2 : WAVES * 64 * D;
+  const long bsS = (long)H * S * D, hsS = (long)S * D, rsS = D;
   hipLaunchKernelGGL((attn_fwd_kernel<D>), grid, dim3(ATT_BLOCK), lds, 0,
-                     q, k, v, slopes, o, lse, S, H, 1);
+                     q, k, v, slopes, o, lse, S, H, 1, bsS, hsS, rsS, bsS,
+                     hsS, rsS);
   CHECK(hipDeviceSynchronize());
   std::vector<__bf16> ho(n);
   hipMemcpy(ho.data(), o, n * 2, hipMemcpyDeviceToHost);
