@@ -279,6 +279,15 @@ class FedServer:
                     f"metrics/eval/LanguageCrossEntropy_client_{i}": l
                     for i, l in enumerate(losses)
                 }
+            # config-gated ICL / gauntlet evaluation on the global model
+            icl_cfg = self.cfg.get("icl_tasks_config") or {}
+            if icl_cfg.get("icl_tasks"):
+                from ..centralised_train import run_icl_eval
+
+                self.layout.copy_to_model(self.client.model)
+                extra.update(
+                    run_icl_eval(self.cfg, self.client.model, self.device)
+                )
             self.history.add_metrics_distributed(
                 server_round,
                 {"metrics/eval/LanguageCrossEntropy_avg": avg, **extra},
