@@ -307,17 +307,30 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
         for (int r = 0; r < 16; ++r) p[r] = s_acc[r];
         (void)need_mask; (void)abase; (void)tile_max;
 #else
+        // The mask branch is specialized at the LOOP level: leaving the
+        // per-element `if (need_mask)` inside the loop gets if-converted
+        // into ~170 always-executed cmp/cndmask instructions per tile
+        // (measured: 860-instruction main loop for 16 MFMAs).
+        if (need_mask) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-          if (need_mask) {
+          for (int r = 0; r < 16; ++r) {
+            const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
             const int key = kv0s + pat;
             const bool masked = (key >= S) || (causal && key > my_q);
             sv = masked ? -1e30f : sv;
+            p[r] = sv;
+            tile_max = fmaxf(tile_max, sv);
           }
-          p[r] = sv;
-          tile_max = fmaxf(tile_max, sv);
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            const float sv =
+                fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
+            p[r] = sv;
+            tile_max = fmaxf(tile_max, sv);
+          }
         }
         tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
         // T13 defer-max: rescale only if some lane's max grew by > THR.
@@ -530,17 +543,28 @@ void attn_bwd_dq_kernel(
           (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
       const float abase = slope * (float)(kv0s - my_q);
       float ds[16];
+      // loop-level mask specialization (cf. fwd): the embedded per-element
+      // if gets if-converted into always-on cmp/cndmask chains.
+      if (need_mask) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-        if (need_mask) {
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
           const int key = kv0s + pat;
           const bool masked = (key >= S) || (causal && key > my_q);
           sv = masked ? -1e30f : sv;
+          const float pv = __expf(sv - my_lse);
+          ds[r] = pv * (dp_acc[r] - my_delta) * scale;
         }
-        const float pv = __expf(sv - my_lse);
-        ds[r] = pv * (dp_acc[r] - my_delta) * scale;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const float sv =
+              fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
+          const float pv = __expf(sv - my_lse);
+          ds[r] = pv * (dp_acc[r] - my_delta) * scale;
+        }
       }
       const int tj = lane & 15;
       const int tg1 = (lane >> 4) & 1;
@@ -733,20 +757,32 @@ void attn_bwd_dkdv_kernel(
           (qt0s + 32 > S) || (causal && block_k_max >= qt0s);
       const float abase = slope * (float)(my_key - qt0s);
       float p[16], ds[16];
+      // loop-level mask specialization (cf. fwd)
+      if (need_mask) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const float l = lse_buf[buf * QTF + sub * 32 + pat];
-        const float dlt = del_buf[buf * QTF + sub * 32 + pat];
-        float sv = fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
-        if (need_mask) {
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const float l = lse_buf[buf * QTF + sub * 32 + pat];
+          const float dlt = del_buf[buf * QTF + sub * 32 + pat];
+          float sv = fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
           const int qi = qt0s + pat;
           const bool masked =
               (my_key >= S) || (causal && my_key > qi) || (qi >= S);
           sv = masked ? -1e30f : sv;
+          p[r] = __expf(sv - l);
+          ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
         }
-        p[r] = __expf(sv - l);
-        ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const float l = lse_buf[buf * QTF + sub * 32 + pat];
+          const float dlt = del_buf[buf * QTF + sub * 32 + pat];
+          const float sv =
+              fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
+          p[r] = __expf(sv - l);
+          ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
+        }
       }
       const int tj = lane & 15;
       const int tg1 = (lane >> 4) & 1;
