@@ -41,6 +41,7 @@ from .runtime import Comm, assign_clients_to_ranks, sample_clients
 from .params_ops import join_payload, split_payload
 from .server_ckpt import (
     copy_old_checkpoints_to_new_run,
+    delete_rounds,
     interpret_resume_round,
     obtain_sorted_rounds,
     resume_from_round,
@@ -318,5 +319,13 @@ class FedServer:
                     {cid: vars(st) for cid, st in self.client.client_states.items()},
                     self.server_steps_cumulative,
                 )
+                # per-round retention (reference cleanup_checkpoints_per_round,
+                # server_app.py:403-405): keep the newest N complete rounds
+                keep = int(self.cfg["photon"].get("save_num_rounds_to_keep", 0) or 0)
+                if keep > 0:
+                    rounds = obtain_sorted_rounds(
+                        self.saving_path, self.run_uuid, self.strategy.state_keys
+                    )
+                    delete_rounds(self.saving_path, self.run_uuid, rounds[-keep:])
             self.comm.barrier()
         return self.history

@@ -232,3 +232,18 @@ def test_fed_training_reduces_loss(tiny_cfg):
         first = first if first is not None else loss
         last = loss
     assert last < first - 0.03, (first, last)
+
+
+def test_round_checkpoint_retention(tiny_cfg, tmp_path):
+    import copy
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    cfg["photon"]["save_num_rounds_to_keep"] = 2
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.run(3)
+    from photon_amd.fed.server_ckpt import obtain_sorted_rounds
+
+    rounds = obtain_sorted_rounds(tmp_path, cfg["run_uuid"], srv.strategy.state_keys)
+    assert rounds == [2, 3]
