@@ -260,6 +260,30 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   stage_load(0);
   stage_write(0);
 
+  // Hoisted LDS read addresses (tile-loop invariant; the swizzle math was
+  // otherwise recomputed per read, ~60 VALU per tile). Offsets BELOW the
+  // swizzle's XOR bits (4..7) cannot be folded as adds, so each distinct
+  // low-offset read gets its own precomputed address; the sub (+4096 B),
+  // s16 (+2048 B) and buf (+2*IMG) offsets sit above bit 7 with no carry
+  // from the low fields and fold into the ds-instruction immediate.
+  const int pv_tj = lane & 15;
+  const int pv_tg1 = (lane >> 4) & 1;
+  unsigned qk_addr[D / 16];  // per-kk QK fragment address (sub=0, buf=0)
+#pragma unroll
+  for (int kk = 0; kk < D / 16; ++kk) {
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+  }
+  unsigned pv_addr[2][D / 32];  // per (rd, db) tr-read address
+#pragma unroll
+  for (int rd = 0; rd < 2; ++rd) {
+#pragma unroll
+    for (int db = 0; db < D / 32; ++db) {
+      const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
+      const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
+      pv_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+    }
+  }
+
   for (int t = 0; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int kv0 = t * KBF;
@@ -364,21 +388,20 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
         // permuted within each 16-row half (pi: row 16*g + j holds
         // dh 16*g + 4*(j&3) + (j>>2)); the epilogue un-permutes.
 #ifndef ABENCH_NO_PV
-        const int tj = lane & 15;         // position within 16-lane group
-        const int tg1 = (lane >> 4) & 1;  // dh half of the 32-row frag
 #pragma unroll
         for (int s16 = 0; s16 < 2; ++s16) {
           bf16x8 pfrag = pack_bfrag(p, 8 * s16);
-          const int koff = sub * 32 + s16 * 16 + 8 * hi + (tj >> 2);
 #pragma unroll
           for (int db = 0; db < D / 32; ++db) {
-            const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
             union { bf16x4 h[2]; bf16x8 v8; } a;
 #pragma unroll
             for (int rd = 0; rd < 2; ++rd) {
-              const int key = koff + 4 * rd;
+              // hoisted: buf/sub/s16 offsets are immediates (the XOR
+              // field key&15 is invariant to +32/+16 key steps)
               a.h[rd] = lds_tr16(
-                  v_img(buf), swz((unsigned)(key * (D * 2) + dhc * 2), key));
+                  v_img(0),
+                  pv_addr[rd][db] + buf * 2 * IMG + sub * 32 * (D * 2) +
+                      s16 * 16 * (D * 2));
             }
             o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a.v8, pfrag, o_acc[db], 0, 0, 0);
@@ -514,6 +537,26 @@ void attn_bwd_dq_kernel(
   stage_load(0);
   stage_write(0);
 
+  // Hoisted LDS read addresses (cf. fwd): same k/v row-image layout, so
+  // one address set serves both images (v_img = k_img + IMG2 immediate).
+  const int pv_tj = lane & 15;
+  const int pv_tg1 = (lane >> 4) & 1;
+  unsigned qk_addr[D / 16];
+#pragma unroll
+  for (int kk = 0; kk < D / 16; ++kk) {
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+  }
+  unsigned tr_addr[2][D / 32];
+#pragma unroll
+  for (int rd = 0; rd < 2; ++rd) {
+#pragma unroll
+    for (int db = 0; db < D / 32; ++db) {
+      const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
+      const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
+      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+    }
+  }
+
   for (int t = 0; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int kv0 = t * KBQ;
@@ -527,10 +570,10 @@ void attn_bwd_dq_kernel(
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
 #pragma unroll
       for (int kk = 0; kk < D / 16; ++kk) {
-        bf16x8 ka =
-            lds_frag(k_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
-        bf16x8 va =
-            lds_frag(v_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
+        const char* base = (const char*)k_img(0) + qk_addr[kk] +
+                           buf * 2 * IMG2 + sub * 32 * (D * 2);
+        bf16x8 ka = *(const bf16x8*)base;
+        bf16x8 va = *(const bf16x8*)(base + IMG2);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[kk], s_acc,
                                                         0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[kk],
@@ -566,21 +609,17 @@ void attn_bwd_dq_kernel(
           ds[r] = pv * (dp_acc[r] - my_delta) * scale;
         }
       }
-      const int tj = lane & 15;
-      const int tg1 = (lane >> 4) & 1;
 #pragma unroll
       for (int s16 = 0; s16 < 2; ++s16) {
         bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
-        const int koff = sub * 32 + s16 * 16 + 8 * hi + (tj >> 2);
 #pragma unroll
         for (int db = 0; db < D / 32; ++db) {
-          const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
           union { bf16x4 h[2]; bf16x8 v8; } a;
 #pragma unroll
           for (int rd = 0; rd < 2; ++rd) {
-            const int key = koff + 4 * rd;
             a.h[rd] = lds_tr16(
-                k_img(buf), swz((unsigned)(key * (D * 2) + dhc * 2), key));
+                k_img(0), tr_addr[rd][db] + buf * 2 * IMG2 +
+                              sub * 32 * (D * 2) + s16 * 16 * (D * 2));
           }
           dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a.v8, dsfrag, dq_acc[db], 0, 0, 0);
@@ -727,6 +766,25 @@ void attn_bwd_dkdv_kernel(
   stage_load(t0);
   stage_write(t0 & 1);
 
+  // Hoisted LDS read addresses (cf. fwd); q and do share the layout.
+  const int pv_tj = lane & 15;
+  const int pv_tg1 = (lane >> 4) & 1;
+  unsigned qk_addr[D / 16];
+#pragma unroll
+  for (int kk = 0; kk < D / 16; ++kk) {
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+  }
+  unsigned tr_addr[2][D / 32];
+#pragma unroll
+  for (int rd = 0; rd < 2; ++rd) {
+#pragma unroll
+    for (int db = 0; db < D / 32; ++db) {
+      const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
+      const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
+      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+    }
+  }
+
   for (int t = t0; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int qt0 = t * QTF;
@@ -742,10 +800,10 @@ void attn_bwd_dkdv_kernel(
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
 #pragma unroll
       for (int kk = 0; kk < D / 16; ++kk) {
-        bf16x8 qa =
-            lds_frag(q_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
-        bf16x8 doa =
-            lds_frag(do_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
+        const char* base = (const char*)q_img(0) + qk_addr[kk] +
+                           buf * 2 * IMG2 + sub * 32 * (D * 2);
+        bf16x8 qa = *(const bf16x8*)base;
+        bf16x8 doa = *(const bf16x8*)(base + IMG2);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[kk], s_acc,
                                                         0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vfrag[kk],
@@ -784,24 +842,20 @@ void attn_bwd_dkdv_kernel(
           ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
         }
       }
-      const int tj = lane & 15;
-      const int tg1 = (lane >> 4) & 1;
 #pragma unroll
       for (int s16 = 0; s16 < 2; ++s16) {
         bf16x8 pfrag = pack_bfrag(p, 8 * s16);
         bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
-        const int qoff = sub * 32 + s16 * 16 + 8 * hi + (tj >> 2);
 #pragma unroll
         for (int db = 0; db < D / 32; ++db) {
-          const int dhc = db * 32 + 16 * tg1 + 4 * (tj & 3);
           union { bf16x4 h[2]; bf16x8 v8; } doa, qa;
 #pragma unroll
           for (int rd = 0; rd < 2; ++rd) {
-            const int qr = qoff + 4 * rd;
-            const unsigned byte =
-                swz((unsigned)(qr * (D * 2) + dhc * 2), qr);
-            doa.h[rd] = lds_tr16(do_img(buf), byte);
-            qa.h[rd] = lds_tr16(q_img(buf), byte);
+            const unsigned byte = tr_addr[rd][db] + buf * 2 * IMG2 +
+                                  sub * 32 * (D * 2) + s16 * 16 * (D * 2);
+            qa.h[rd] = lds_tr16(q_img(0), byte);
+            doa.h[rd] = lds_tr16((const __bf16*)((const char*)q_img(0) + IMG2),
+                                 byte);
           }
           dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               doa.v8, pfrag, dv_acc[db], 0, 0, 0);
