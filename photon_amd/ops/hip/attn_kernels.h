@@ -584,7 +584,10 @@ void attn_bwd_dq_kernel(
       const bool need_mask =
           (kv0s + 32 > S) ||
           (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
-      const float abase = slope * (float)(kv0s - my_q);
+      // abase folds ln(scale): exp yields p*scale, so ds skips the
+      // per-element multiply (cf. dkdv).
+      const float abase =
+          fmaf(slope, (float)(kv0s - my_q), __logf(scale));
       float ds[16];
       // loop-level mask specialization (cf. fwd): the embedded per-element
       // if gets if-converted into always-on cmp/cndmask chains.
@@ -597,7 +600,7 @@ void attn_bwd_dq_kernel(
           const bool masked = (key >= S) || (causal && key > my_q);
           sv = masked ? -1e30f : sv;
           const float pv = __expf(sv - my_lse);
-          ds[r] = pv * (dp_acc[r] - my_delta) * scale;
+          ds[r] = pv * (dp_acc[r] - my_delta);
         }
       } else {
 #pragma unroll
@@ -606,7 +609,7 @@ void attn_bwd_dq_kernel(
           const float sv =
               fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
           const float pv = __expf(sv - my_lse);
-          ds[r] = pv * (dp_acc[r] - my_delta) * scale;
+          ds[r] = pv * (dp_acc[r] - my_delta);
         }
       }
 #pragma unroll
@@ -691,6 +694,7 @@ void attn_bwd_dkdv_kernel(
   const float scale = rsqrtf((float)D);
   const long ibase = (bh / H) * bs_i + (long)h * hs_i;
   const long obase = (bh / H) * bs_o + (long)h * hs_o;
+  const float log_scale = __logf(scale);
   const int k0 = blockIdx.x * (WAVES * KB) + wave * KB;
   const int my_key = k0 + lq;
 
@@ -813,7 +817,14 @@ void attn_bwd_dkdv_kernel(
       // need_mask is BLOCK-uniform (scalar branch).
       const bool need_mask =
           (qt0s + 32 > S) || (causal && block_k_max >= qt0s);
-      const float abase = slope * (float)(my_key - qt0s);
+      // abase folds BOTH the ALiBi row constant and ln(scale): the exp
+      // then yields p*scale directly — ds needs no per-element *scale and
+      // dv_acc (which wants raw p) is divided by scale once in the
+      // epilogue. (Vectorized lse/del b128 loads were tried and REVERTED:
+      // +32 live VGPRs pushed dkdv<64> to 256+80B-scratch spilling, a net
+      // 30% slowdown.)
+      const float abase =
+          fmaf(slope, (float)(my_key - qt0s), log_scale);
       float p[16], ds[16];
       // loop-level mask specialization (cf. fwd)
       if (need_mask) {
@@ -827,8 +838,8 @@ void attn_bwd_dkdv_kernel(
           const bool masked =
               (my_key >= S) || (causal && my_key > qi) || (qi >= S);
           sv = masked ? -1e30f : sv;
-          p[r] = __expf(sv - l);
-          ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
+          p[r] = __expf(sv - l);  // == p_raw * scale
+          ds[r] = p[r] * (dp_acc[r] - dlt);
         }
       } else {
 #pragma unroll
@@ -838,8 +849,8 @@ void attn_bwd_dkdv_kernel(
           const float dlt = del_buf[buf * QTF + sub * 32 + pat];
           const float sv =
               fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
-          p[r] = __expf(sv - l);
-          ds[r] = p[r] * (dp_acc[r] - dlt) * scale;
+          p[r] = __expf(sv - l);  // == p_raw * scale
+          ds[r] = p[r] * (dp_acc[r] - dlt);
         }
       }
 #pragma unroll
@@ -867,6 +878,16 @@ void attn_bwd_dkdv_kernel(
     if (t + 1 < n_tiles) stage_write(1 - buf);
   }
   __syncthreads();  // protect epilogue smem reuse
+
+  // dv accumulated p*scale (see abase fold): undo once here.
+  {
+    const float inv_scale = 1.f / scale;
+#pragma unroll
+    for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) dv_acc[db][r] *= inv_scale;
+    }
+  }
 
   // epilogue: two bounces (dk then dv) through per-wave LDS
   // (tr-read rows are natural order; see fwd epilogue note)
