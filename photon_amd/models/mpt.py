@@ -16,9 +16,7 @@ fallback, photon_llm_125M.sh:121).
 
 from __future__ import annotations
 
-import math
-from dataclasses import dataclass, field
-from typing import Any
+from dataclasses import dataclass
 
 import torch
 import torch.nn as nn

@@ -189,6 +189,16 @@ class Trainer:
         folder = Path(folder or self.save_folder or ".")
         folder.mkdir(parents=True, exist_ok=True)
         path = folder / self.checkpoint_name()
+        # retention: save_num_checkpoints_to_keep (Composer semantics;
+        # <= 0 keeps everything)
+        keep = int(self.cfg.get("save_num_checkpoints_to_keep", -1) or -1)
+        if keep > 0:
+            existing = sorted(
+                folder.glob(f"ep*-ba*-rank{self.rank}.pt"),
+                key=lambda p: p.stat().st_mtime,
+            )
+            for old in existing[: max(0, len(existing) - (keep - 1))]:
+                old.unlink(missing_ok=True)
         state = {
             "state": {
                 "model": self.model.state_dict(),

@@ -200,3 +200,15 @@ def test_monitors_and_profiler(tiny_llm_config, tmp_path):
     assert any(k.startswith("lr-") for k in keys)
     assert "optimizer/l2_norm_grad" in keys
     assert list((tmp_path / "traces").glob("*.json*")), "chrome trace written"
+
+
+def test_checkpoint_retention(tiny_llm_config, tmp_path):
+    cfg = dict(tiny_llm_config)
+    cfg["save_num_checkpoints_to_keep"] = 2
+    tr = make_trainer(cfg, tmp_path)
+    for _ in range(4):
+        tr.fit("1ba")
+        tr.save_checkpoint()
+    kept = sorted(tmp_path.glob("ep*-ba*-rank0.pt"))
+    assert len(kept) == 2
+    assert kept[-1].name == "ep0-ba4-rank0.pt"
