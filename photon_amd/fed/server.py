@@ -226,6 +226,17 @@ class FedServer:
             self.client_m1.copy_(m1_avg)
             self.client_m2.copy_(m2_avg)
 
+        # debug mode (photon.debug_checks): runtime invariants in the spirit
+        # of the reference's parameters_checker asserts —
+        # replicated state must stay bit-identical across ranks.
+        if self.cfg["photon"].get("debug_checks", False):
+            check = self.strategy.params.clone()
+            self.comm.broadcast_flat(check, src=0)
+            from .params_ops import parameters_checker
+
+            parameters_checker(check, self.strategy.params, equal=True)
+            assert torch.isfinite(fedavg_flat).all(), "non-finite aggregate"
+
         # replicated server-opt update
         strat_metrics = self.strategy.update(
             fedavg_flat, server_round, len(sampled)

@@ -196,7 +196,7 @@ def test_clip_kernel_vs_torch(dev, ext):
 # Flash attention
 # ---------------------------------------------------------------------------
 @pytest.mark.parametrize("D", [64, 128])
-@pytest.mark.parametrize("S", [128, 256, 2048])
+@pytest.mark.parametrize("S", [128, 256, 300, 2048])
 def test_attn_fwd_vs_fp32_ref(dev, ext, D, S):
     from photon_amd.ops.attention import alibi_slopes, reference_attention_fp32
 
@@ -229,7 +229,7 @@ def test_attn_fwd_noncausal(dev, ext):
 
 
 @pytest.mark.parametrize("D", [64, 128])
-@pytest.mark.parametrize("S", [128, 512])
+@pytest.mark.parametrize("S", [128, 300, 512])
 def test_attn_bwd_vs_fp32_ref(dev, ext, D, S):
     from photon_amd.ops.attention import alibi_slopes, reference_attention_fp32
 
