@@ -39,6 +39,8 @@ def main() -> None:
     ap.add_argument("--global-batch", type=int, default=256)
     ap.add_argument("--microbatch", type=int, default=32)
     ap.add_argument("--attn", default="flash", choices=["flash", "torch"])
+    ap.add_argument("--hip-graphs", action="store_true",
+                    help="capture the microbatch fwd+bwd in a hipGraph")
     args = ap.parse_args()
 
     from photon_amd.conf import compose, config_yaml_dir
@@ -68,6 +70,8 @@ def main() -> None:
     llm["global_train_batch_size"] = args.global_batch
     llm["device_train_microbatch_size"] = args.microbatch
     llm["precision"] = "amp_bf16" if use_cuda else "fp32"
+    if args.hip_graphs:
+        llm["use_hip_graphs"] = True
 
     torch.manual_seed(17)
     model = build_model(llm)

@@ -76,6 +76,9 @@ class Trainer:
         )
         self.max_duration = duration_to_batches(llm_config.get("max_duration", "1000000ba"))
         self.metrics: dict[str, float] = {}
+        # hipGraph-captured microbatch (train/graphs.py); opt-in, GPU only
+        self.use_hip_graphs = bool(llm_config.get("use_hip_graphs", False))
+        self._graphed = None
         # reference default callback set (speed/lr/memory/runtime/optimizer
         # monitors, mpt-125m.yaml:98-109)
         self.monitors = build_monitors(llm_config.get("callbacks"))
@@ -95,24 +98,43 @@ class Trainer:
     # -- train --------------------------------------------------------------
     def train_batch(self, batches: list[dict]) -> float:
         """One optimization batch = grad_accum microbatches. Returns loss."""
-        self.optimizer.zero_grad(set_to_none=True)
         n = len(batches)
         total_loss_t = None
-        # ONE autocast region for the whole accumulation loop: the autocast
-        # weight-cast cache then converts each fp32 weight to bf16 once per
-        # optimization batch instead of once per microbatch (measured ~8% of
-        # step time in cast/copy kernels at grad_accum=16, profiles/r01).
-        with self.autocast():
+        if self.use_hip_graphs and self.device.type == "cuda":
+            # graph path: grads are captured tensors — zero in place
+            if self._graphed is None:
+                from .graphs import GraphedMicrobatch
+
+                ids0 = batches[0]["input_ids"]
+                self._graphed = GraphedMicrobatch(
+                    self.model, self.autocast, float(self.grad_accum),
+                    ids0.shape[0], ids0.shape[1], self.device,
+                )
+            else:
+                self._graphed._zero_grads()
             for mb in batches:
                 ids = mb["input_ids"].to(self.device, non_blocking=True)
-                out = self.model(ids, labels=ids)
-                loss = out["loss"] / n
-                loss.backward()
+                loss = self._graphed.run(ids)
                 total_loss_t = (
-                    loss.detach()
-                    if total_loss_t is None
-                    else total_loss_t + loss.detach()
+                    loss.clone() if total_loss_t is None else total_loss_t + loss
                 )
+        else:
+            self.optimizer.zero_grad(set_to_none=True)
+            # ONE autocast region for the whole accumulation loop: the
+            # autocast weight-cast cache then converts each fp32 weight to
+            # bf16 once per optimization batch instead of once per
+            # microbatch (measured ~8% of step time at grad_accum=16).
+            with self.autocast():
+                for mb in batches:
+                    ids = mb["input_ids"].to(self.device, non_blocking=True)
+                    out = self.model(ids, labels=ids)
+                    loss = out["loss"] / n
+                    loss.backward()
+                    total_loss_t = (
+                        loss.detach()
+                        if total_loss_t is None
+                        else total_loss_t + loss.detach()
+                    )
         if self.grad_sync_hook is not None:
             self.grad_sync_hook(self.model)
         if self.clip_norm > 0:
