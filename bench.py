@@ -41,6 +41,8 @@ def main() -> None:
     ap.add_argument("--attn", default="flash", choices=["flash", "torch"])
     ap.add_argument("--hip-graphs", action="store_true",
                     help="capture the microbatch fwd+bwd in a hipGraph")
+    ap.add_argument("--master-weights", action="store_true",
+                    help="bf16 model weights + fp32 optimizer masters")
     args = ap.parse_args()
 
     from photon_amd.conf import compose, config_yaml_dir
@@ -72,6 +74,8 @@ def main() -> None:
     llm["precision"] = "amp_bf16" if use_cuda else "fp32"
     if args.hip_graphs:
         llm["use_hip_graphs"] = True
+    if args.master_weights:
+        llm["master_weights"] = True
 
     torch.manual_seed(17)
     model = build_model(llm)

@@ -149,12 +149,18 @@ class FedClient:
         global_flat, m1_in, m2_in = manipulate_pre_training(
             payload, layout, fl, cid, local_params=self._personal.get(cid)
         )
-        # set params from the global buffer (HBM->HBM copies, no host hop)
+        # set params from the global buffer (HBM->HBM copies, no host hop);
+        # with bf16 weights the fp32 optimizer masters get the EXACT global
+        # values (no bf16 round-trip)
         views = layout.layer_views_of(global_flat)
         params = dict(self.model.named_parameters())
         with torch.no_grad():
-            for n, v in zip(layout.names, views):
-                params[n].data.copy_(v.to(params[n].dtype))
+            if getattr(trainer, "master_weights", False):
+                order = [params[n] for n in layout.names]
+                trainer.optimizer.sync_masters(order, views)
+            else:
+                for n, v in zip(layout.names, views):
+                    params[n].data.copy_(v.to(params[n].dtype))
         set_params_time = time.time() - t0
 
         if reset_optimizer:

@@ -18,11 +18,13 @@ torch::Tensor ce_fwd_bwd_inplace(torch::Tensor logits, torch::Tensor targets);
 void adamw_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                 double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2);
+                double bc1, double bc2,
+                std::vector<torch::Tensor> masters);
 void adopt_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                 double lr, double b1, double b2, double eps, double wd,
-                double clip, long step);
+                double clip, long step,
+                std::vector<torch::Tensor> masters);
 torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> gs);
 void multi_tensor_scale_clip(std::vector<torch::Tensor> gs,
                              torch::Tensor total_norm, double max_norm);

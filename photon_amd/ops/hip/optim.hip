@@ -21,6 +21,7 @@ struct TensorBatch {
   T* g[MT_CHUNK];
   float* m[MT_CHUNK];
   float* v[MT_CHUNK];
+  float* w[MT_CHUNK];  // fp32 master weights (null = p IS the master)
   long n[MT_CHUNK];
   int count;
 };
@@ -39,15 +40,17 @@ __global__ void adamw_kernel(TensorBatch<T> batch, float lr, float b1,
   T* g = batch.g[t];
   float* m = batch.m[t];
   float* v = batch.v[t];
+  float* w = batch.w[t];  // fp32 master (PURE mixed precision), or null
   const long n = batch.n[t];
   const float decay = 1.f - lr * wd;
   for (long i = (long)(bi * BLOCK + threadIdx.x); i < n; i += (long)nb * BLOCK) {
     float gf = load_f32<T>(g, i);
     float mi = m[i] = b1 * m[i] + (1.f - b1) * gf;
     float vi = v[i] = b2 * v[i] + (1.f - b2) * gf * gf;
-    float pv = load_f32<T>(p, i);
+    float pv = w ? w[i] : load_f32<T>(p, i);
     if (wd != 0.f) pv *= decay;
     pv -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    if (w) w[i] = pv;
     store_f32<T>(p, i, pv);
   }
 }
@@ -63,6 +66,7 @@ __global__ void adopt_kernel(TensorBatch<T> batch, float lr, float b1,
   T* g = batch.g[t];
   float* m = batch.m[t];
   float* v = batch.v[t];
+  float* w = batch.w[t];
   const long n = batch.n[t];
   const float decay = 1.f - lr * wd;
   for (long i = (long)(bi * BLOCK + threadIdx.x); i < n; i += (long)nb * BLOCK) {
@@ -74,9 +78,10 @@ __global__ void adopt_kernel(TensorBatch<T> batch, float lr, float b1,
     float c = gf / fmaxf(sqrtf(v[i]), eps);
     c = fminf(fmaxf(c, -clip), clip);
     float mi = m[i] = b1 * m[i] + (1.f - b1) * c;
-    float pv = load_f32<T>(p, i);
+    float pv = w ? w[i] : load_f32<T>(p, i);
     if (wd != 0.f) pv *= decay;
     pv -= lr * mi;
+    if (w) w[i] = pv;
     store_f32<T>(p, i, pv);
     v[i] = b2 * v[i] + (1.f - b2) * gf * gf;
   }
@@ -146,7 +151,8 @@ TensorBatch<T> make_batch(const std::vector<torch::Tensor>* ps,
                           const std::vector<torch::Tensor>& gs,
                           const std::vector<torch::Tensor>* ms,
                           const std::vector<torch::Tensor>* vs, size_t start,
-                          size_t count) {
+                          size_t count,
+                          const std::vector<torch::Tensor>* ws = nullptr) {
   TensorBatch<T> b{};
   b.count = (int)count;
   for (size_t i = 0; i < count; ++i) {
@@ -154,6 +160,8 @@ TensorBatch<T> make_batch(const std::vector<torch::Tensor>* ps,
     b.g[i] = (T*)gs[start + i].data_ptr();
     b.m[i] = ms ? (*ms)[start + i].data_ptr<float>() : nullptr;
     b.v[i] = vs ? (*vs)[start + i].data_ptr<float>() : nullptr;
+    b.w[i] = (ws && !ws->empty()) ? (*ws)[start + i].data_ptr<float>()
+                                  : nullptr;
     b.n[i] = gs[start + i].numel();
   }
   return b;
@@ -163,12 +171,13 @@ TensorBatch<T> make_batch(const std::vector<torch::Tensor>* ps,
 void adamw_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                 double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2) {
+                double bc1, double bc2,
+                std::vector<torch::Tensor> masters = {}) {
   const int GRID = 1024;
   for (size_t s = 0; s < ps.size(); s += MT_CHUNK) {
     size_t c = std::min((size_t)MT_CHUNK, ps.size() - s);
     DISPATCH_DTYPE(ps[s], "adamw_step", {
-      auto batch = make_batch<scalar_t>(&ps, gs, &ms, &vs, s, c);
+      auto batch = make_batch<scalar_t>(&ps, gs, &ms, &vs, s, c, &masters);
       hipLaunchKernelGGL((adamw_kernel<scalar_t, OPT_BLOCK>), dim3(GRID),
                          dim3(OPT_BLOCK), 0, cur_stream(), batch, (float)lr,
                          (float)b1, (float)b2, (float)eps, (float)wd,
@@ -180,12 +189,13 @@ void adamw_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
 void adopt_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
                 double lr, double b1, double b2, double eps, double wd,
-                double clip, long step) {
+                double clip, long step,
+                std::vector<torch::Tensor> masters = {}) {
   const int GRID = 1024;
   for (size_t s = 0; s < ps.size(); s += MT_CHUNK) {
     size_t c = std::min((size_t)MT_CHUNK, ps.size() - s);
     DISPATCH_DTYPE(ps[s], "adopt_step", {
-      auto batch = make_batch<scalar_t>(&ps, gs, &ms, &vs, s, c);
+      auto batch = make_batch<scalar_t>(&ps, gs, &ms, &vs, s, c, &masters);
       hipLaunchKernelGGL((adopt_kernel<scalar_t, OPT_BLOCK>), dim3(GRID),
                          dim3(OPT_BLOCK), 0, cur_stream(), batch, (float)lr,
                          (float)b1, (float)b2, (float)eps, (float)wd,

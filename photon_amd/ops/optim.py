@@ -30,8 +30,11 @@ class _FusedStepMixin:
     """Shared fused multi-tensor step plumbing."""
 
     def _bucket(self, group):
-        """Return (params, grads, m, v, steps) lists for params with grads."""
-        ps, gs, ms, vs = [], [], [], []
+        """Return (params, grads, m, v, masters) lists for params with
+        grads. Non-fp32 params get an fp32 MASTER copy (PURE mixed
+        precision: bf16 weights in the model, exact update state here)."""
+        ps, gs, ms, vs, ws = [], [], [], [], []
+        any_master = False
         for p in group["params"]:
             if p.grad is None:
                 continue
@@ -40,11 +43,21 @@ class _FusedStepMixin:
                 state["step"] = 0
                 state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                 state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+            if p.dtype != torch.float32 and "master" not in state:
+                state["master"] = p.detach().to(torch.float32).clone()
             ps.append(p)
             gs.append(p.grad)
             ms.append(state["exp_avg"])
             vs.append(state["exp_avg_sq"])
-        return ps, gs, ms, vs
+            master = state.get("master")
+            ws.append(master if master is not None else None)
+            any_master = any_master or master is not None
+        if any_master:
+            # HIP path needs a uniform list; fp32 params use themselves
+            ws = [w if w is not None else p.detach() for w, p in zip(ws, ps)]
+        else:
+            ws = []
+        return ps, gs, ms, vs, ws
 
     def _bump_steps(self, ps):
         for p in ps:
@@ -60,6 +73,18 @@ class _FusedStepMixin:
             m1.append(st.get("exp_avg", torch.zeros_like(p, dtype=torch.float32)))
             m2.append(st.get("exp_avg_sq", torch.zeros_like(p, dtype=torch.float32)))
         return m1, m2
+
+    def sync_masters(self, order: list[torch.nn.Parameter],
+                     fp32_views: list[torch.Tensor]) -> None:
+        """Overwrite fp32 masters (and the bf16 params) with exact fp32
+        values — the federated round's parameter set must not round-trip
+        through bf16 before entering the master copy."""
+        for p, src in zip(order, fp32_views):
+            st = self.state.setdefault(p, {})
+            if p.dtype != torch.float32:
+                st["master"] = src.detach().to(p.device, torch.float32) \
+                    .view_as(p).clone()
+            p.data.copy_(src.view_as(p).to(p.dtype))
 
     def import_momenta(
         self,
@@ -92,7 +117,7 @@ class DecoupledAdamW(Optimizer, _FusedStepMixin):
     def step(self, closure=None):
         loss = closure() if closure is not None else None
         for group in self.param_groups:
-            ps, gs, ms, vs = self._bucket(group)
+            ps, gs, ms, vs, ws = self._bucket(group)
             if not ps:
                 continue
             step = self._bump_steps(ps)
@@ -102,16 +127,22 @@ class DecoupledAdamW(Optimizer, _FusedStepMixin):
             bc1 = 1 - b1**step
             bc2 = 1 - b2**step
             if use_hip(ps[0]):
-                hip_ext().adamw_step(ps, gs, ms, vs, lr, b1, b2, eps, wd, bc1, bc2)
+                hip_ext().adamw_step(ps, gs, ms, vs, lr, b1, b2, eps, wd,
+                                     bc1, bc2, ws)
                 continue
-            for p, g, m, v in zip(ps, gs, ms, vs):
+            for i, (p, g, m, v) in enumerate(zip(ps, gs, ms, vs)):
                 gf = g.float()
                 m.mul_(b1).add_(gf, alpha=1 - b1)
                 v.mul_(b2).addcmul_(gf, gf, value=1 - b2)
                 denom = (v / bc2).sqrt_().add_(eps)
+                target = ws[i] if ws else p
                 if wd != 0.0:
-                    p.mul_(1 - lr * wd)
-                p.addcdiv_((m / bc1).to(p.dtype), denom.to(p.dtype), value=-lr)
+                    target.mul_(1 - lr * wd)
+                if ws:
+                    target.addcdiv_(m / bc1, denom, value=-lr)
+                    p.data.copy_(target.to(p.dtype))
+                else:
+                    p.addcdiv_((m / bc1).to(p.dtype), denom.to(p.dtype), value=-lr)
         return loss
 
 
@@ -135,7 +166,7 @@ class ADOPT(Optimizer, _FusedStepMixin):
     def step(self, closure=None):
         loss = closure() if closure is not None else None
         for group in self.param_groups:
-            ps, gs, ms, vs = self._bucket(group)
+            ps, gs, ms, vs, ws = self._bucket(group)
             if not ps:
                 continue
             step = self._bump_steps(ps)
@@ -144,9 +175,10 @@ class ADOPT(Optimizer, _FusedStepMixin):
             )
             clip = (step - 1) ** 0.25 if step > 1 else 1.0
             if use_hip(ps[0]):
-                hip_ext().adopt_step(ps, gs, ms, vs, lr, b1, b2, eps, wd, float(clip), step)
+                hip_ext().adopt_step(ps, gs, ms, vs, lr, b1, b2, eps, wd,
+                                     float(clip), step, ws)
                 continue
-            for p, g, m, v in zip(ps, gs, ms, vs):
+            for i, (p, g, m, v) in enumerate(zip(ps, gs, ms, vs)):
                 gf = g.float()
                 if step == 1:
                     # v_0 = g_0^2; no parameter update on the first step
@@ -155,9 +187,14 @@ class ADOPT(Optimizer, _FusedStepMixin):
                 c = gf / v.sqrt().clamp_min(eps)
                 c.clamp_(-clip, clip)
                 m.mul_(b1).add_(c, alpha=1 - b1)
+                target = ws[i] if ws else p
                 if wd != 0.0:
-                    p.mul_(1 - lr * wd)
-                p.add_(m.to(p.dtype), alpha=-lr)
+                    target.mul_(1 - lr * wd)
+                if ws:
+                    target.add_(m, alpha=-lr)
+                    p.data.copy_(target.to(p.dtype))
+                else:
+                    p.add_(m.to(p.dtype), alpha=-lr)
                 v.mul_(b2).addcmul_(gf, gf, value=1 - b2)
         return loss
 

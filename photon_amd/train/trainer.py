@@ -53,6 +53,15 @@ class Trainer:
             else ("cuda" if torch.cuda.is_available() else "cpu")
         )
         self.model = model.to(self.device)
+        # PURE mixed precision (reference fsdp_config mixed_precision: PURE):
+        # bf16 weights in the model, fp32 masters inside the optimizer.
+        # Kills the per-step autocast weight-cast kernels and is required
+        # for hipGraphs to beat eager (train/graphs.py).
+        self.master_weights = bool(
+            llm_config.get("master_weights", False)
+        ) and self.device.type == "cuda"
+        if self.master_weights:
+            self.model = self.model.to(torch.bfloat16)
         self.train_loader = train_loader
         self.eval_loader = eval_loader
         self.world_size = world_size

@@ -203,3 +203,42 @@ def test_fed_round_with_momenta(tmp_path):
     server.run_round(1)
     assert float(server.client_m1.abs().sum()) > 0, "momenta aggregated"
     server.run_round(2)
+
+
+def test_master_weights_cpu_matches_fp32_math():
+    """bf16 param + fp32 master (PURE mixed precision): the master follows
+    exact fp32 AdamW math; the bf16 param is its rounded copy."""
+    from photon_amd.ops.optim import DecoupledAdamW
+
+    torch.manual_seed(0)
+    p_bf = torch.nn.Parameter(torch.randn(64, dtype=torch.bfloat16))
+    ref = p_bf.detach().float().clone()
+    m = torch.zeros(64)
+    v = torch.zeros(64)
+    opt = DecoupledAdamW([p_bf], lr=1e-2, betas=(0.9, 0.95), eps=1e-8,
+                         weight_decay=0.01)
+    for step in range(1, 4):
+        g = torch.randn(64, dtype=torch.bfloat16)
+        p_bf.grad = g.clone()
+        opt.step()
+        # reference fp32 math on the same bf16 grads
+        gf = g.float()
+        m.mul_(0.9).add_(gf, alpha=0.1)
+        v.mul_(0.95).addcmul_(gf, gf, value=0.05)
+        bc1, bc2 = 1 - 0.9**step, 1 - 0.95**step
+        ref.mul_(1 - 1e-2 * 0.01)
+        ref.addcdiv_(m / bc1, (v / bc2).sqrt().add_(1e-8), value=-1e-2)
+        master = opt.state[p_bf]["master"]
+        assert torch.allclose(master, ref, atol=1e-6), (master - ref).abs().max()
+        assert torch.equal(p_bf.detach(), master.to(torch.bfloat16))
+
+
+def test_sync_masters_exact_fp32():
+    from photon_amd.ops.optim import DecoupledAdamW
+
+    p_bf = torch.nn.Parameter(torch.randn(8, dtype=torch.bfloat16))
+    opt = DecoupledAdamW([p_bf], lr=1e-2)
+    src = torch.randn(8)  # exact fp32 global params
+    opt.sync_masters([p_bf], [src])
+    assert torch.equal(opt.state[p_bf]["master"], src)
+    assert torch.equal(p_bf.detach(), src.to(torch.bfloat16))
