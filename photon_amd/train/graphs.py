@@ -5,12 +5,12 @@ Captures ONE fixed-shape microbatch forward+backward into a hipGraph
 (torch.cuda.CUDAGraph is hipGraph on ROCm) and replays it per microbatch:
 removes the per-kernel launch overhead of the ~80-kernel microbatch body.
 
-Measured at MPT-125M/gb256/mb32: graphs 601k vs eager 616k tokens/s — the
-replay re-executes the captured fp32->bf16 weight casts every microbatch,
-while the eager path's single-autocast-region cache converts once per
-optimization batch; the launch-overhead win (~1.5%) does not cover that
-until weights are stored bf16 with fp32 masters (NOTES_NEXT_ROUND.md).
-Until then this stays opt-in (llm_config.use_hip_graphs).
+Measured at MPT-125M/gb256/mb32: graphs 601-606k vs eager 611-616k
+tokens/s, both with fp32+autocast and with bf16 master-weight mode — the
+~10-16 us replay floor over 8 microbatches plus graph-pool memory effects
+eat the launch-overhead win at this kernel count. Opt-in
+(llm_config.use_hip_graphs); expected to pay off for smaller models or
+larger grad-accum degrees.
 
 Contract:
 * input shape is fixed (microbatch_size x seq_len) — the bench/fed training
