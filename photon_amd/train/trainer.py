@@ -75,7 +75,9 @@ class Trainer:
         self.timestamp = Timestamp()
 
         self.precision = str(llm_config.get("precision", "amp_bf16"))
-        self.microbatch = int(llm_config.get("device_train_microbatch_size", 8))
+        self.microbatch = self._resolve_microbatch(
+            llm_config.get("device_train_microbatch_size", 8)
+        )
         self.global_batch = int(llm_config.get("global_train_batch_size", 256))
         clip_cfg = (llm_config.get("algorithms") or {}).get("gradient_clipping") or {}
         self.clip_norm = (
@@ -92,6 +94,23 @@ class Trainer:
         # monitors, mpt-125m.yaml:98-109)
         self.monitors = build_monitors(llm_config.get("callbacks"))
         self.profiler_cfg = llm_config.get("profiler")
+
+    def _resolve_microbatch(self, value) -> int:
+        """'auto' support (the reference's device_train_microbatch_size:
+        auto, Composer semantics): a 288-GB-HBM heuristic by model width —
+        measured-safe sizes at seq<=4096 on MI355X (Composer's dynamic OOM
+        backoff is unnecessary at this memory headroom)."""
+        if isinstance(value, str) and value.strip().lower() == "auto":
+            d = 768
+            cfg = getattr(self.model, "cfg", None)
+            if cfg is not None:
+                d = int(getattr(cfg, "d_model", 768))
+            table = [(1024, 32), (2048, 16), (2560, 8), (4096, 8)]
+            for width, mb in table:
+                if d <= width:
+                    return mb
+            return 4
+        return int(value)
 
     # -- precision ----------------------------------------------------------
     def autocast(self):

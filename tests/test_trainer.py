@@ -212,3 +212,10 @@ def test_checkpoint_retention(tiny_llm_config, tmp_path):
     kept = sorted(tmp_path.glob("ep*-ba*-rank0.pt"))
     assert len(kept) == 2
     assert kept[-1].name == "ep0-ba4-rank0.pt"
+
+
+def test_microbatch_auto(tiny_llm_config, tmp_path):
+    cfg = dict(tiny_llm_config)
+    cfg["device_train_microbatch_size"] = "auto"
+    tr = make_trainer(cfg, tmp_path)
+    assert isinstance(tr.microbatch, int) and tr.microbatch >= 1
