@@ -212,7 +212,16 @@ class Trainer:
         if self.eval_loader is None:
             return {}
         self.model.eval()
-        n = subset_num_batches if subset_num_batches > 0 else 8
+        if subset_num_batches > 0:
+            n = subset_num_batches
+        else:
+            # -1 = the full eval split when it is finite (reference
+            # semantics); synthetic/infinite streams get a fixed window
+            try:
+                n = max(len(self.eval_loader.dataset) // self.eval_loader.batch_size, 1)
+                n = min(n, 1000)
+            except TypeError:
+                n = 8
         total, count = 0.0, 0
         for _ in range(n):
             batch = self.eval_loader.next_batch()
