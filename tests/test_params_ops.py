@@ -242,3 +242,32 @@ def test_sync_masters_exact_fp32():
     opt.sync_masters([p_bf], [src])
     assert torch.equal(opt.state[p_bf]["master"], src)
     assert torch.equal(p_bf.detach(), src.to(torch.bfloat16))
+
+
+def test_fed_round_momenta_plus_personalization(tmp_path):
+    """Combination: aggregate_momenta + personalized_layers + partial
+    participation in one round loop (feature interaction check)."""
+    from photon_amd.fed.runtime import Comm
+    from photon_amd.fed.server import FedServer
+
+    cfg = compose(config_yaml_dir(), "base", ["llm_config=mpt-125m"]).to_plain()
+    llm = cfg["llm_config"]
+    llm["model"].update({"d_model": 64, "n_heads": 2, "n_layers": 2,
+                         "max_seq_len": 32, "vocab_size": 128})
+    llm["model"]["attn_config"]["attn_impl"] = "torch"
+    llm.update({"global_train_batch_size": 2, "device_train_microbatch_size": 2,
+                "local_steps": "2ba", "precision": "fp32"})
+    cfg["fl"].update({
+        "n_total_clients": 4, "n_clients_per_round": 2, "n_rounds": 2,
+        "aggregate_momenta": True, "reset_optimizer": False,
+        "personalized_layers": ["norm_f"], "eval_period": 0,
+    })
+    cfg["photon"]["checkpoint"] = False
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    m1 = srv.run_round(1)
+    m2 = srv.run_round(2)
+    assert m1["server/sampled_clients"] == 2
+    assert float(srv.client_m1.abs().sum()) > 0
+    assert torch.isfinite(srv.strategy.params).all()
