@@ -140,6 +140,25 @@ class FedServer:
             )
             self.server_steps_cumulative = int(state.get("server_steps_cumulative", 0))
             self.history.load_state(state.get("history", {}))
+            # client_state round-trips as str() in state.bin (reference
+            # s3_utils.py:374-389); restore steps_done per client so lr
+            # schedules and skip-and-load continue where they left off
+            import ast
+
+            from .client import ClientState
+
+            raw = state.get("client_state")
+            if raw:
+                try:
+                    parsed = ast.literal_eval(raw) if isinstance(raw, str) else raw
+                    for cid, st in parsed.items():
+                        self.client.client_states[int(cid)] = ClientState(
+                            cid=int(cid),
+                            steps_done=int(st.get("steps_done", 0)),
+                            metrics=dict(st.get("metrics", {})),
+                        )
+                except (ValueError, SyntaxError):
+                    pass
             self.start_round = target + 1
             resumed = target
         else:

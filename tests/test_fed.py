@@ -247,3 +247,22 @@ def test_round_checkpoint_retention(tiny_cfg, tmp_path):
 
     rounds = obtain_sorted_rounds(tmp_path, cfg["run_uuid"], srv.strategy.state_keys)
     assert rounds == [2, 3]
+
+
+def test_resume_restores_client_state(tiny_cfg, tmp_path):
+    import copy
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.run(2)
+    steps = {cid: st.steps_done for cid, st in srv.client.client_states.items()}
+    assert steps
+
+    cfg2 = copy.deepcopy(cfg)
+    cfg2["photon"]["resume_round"] = -1
+    srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
+    srv2.initialize()
+    assert {c: s.steps_done for c, s in srv2.client.client_states.items()} == steps
+    assert srv2.server_steps_cumulative == srv.server_steps_cumulative
