@@ -145,7 +145,7 @@ def test_adamw_kernel_vs_torch(dev, ext):
     lr, b1, b2, eps = 1e-3, 0.9, 0.95, 1e-8
     for step in range(1, 4):
         bc1, bc2 = 1 - b1**step, 1 - b2**step
-        ext.adamw_step(ps, gs, ms, vs, lr, b1, b2, eps, 0.0, bc1, bc2)
+        ext.adamw_step(ps, gs, ms, vs, lr, b1, b2, eps, 0.0, bc1, bc2, [])
     # torch reference
     ms_r = [torch.zeros_like(p) for p in ps_ref]
     vs_r = [torch.zeros_like(p) for p in ps_ref]
@@ -359,3 +359,32 @@ def test_layernorm_d2560():
     assert (y.float() - ref).abs().max() < 2e-2
     y.sum().backward()
     assert x.grad is not None and ln.weight.grad is not None
+
+
+@pytest.mark.gpu
+def test_adamw_kernel_master_weights():
+    """Fused AdamW with bf16 params + fp32 masters matches the fp32 path."""
+    from photon_amd.ops import hip_ext
+
+    torch.manual_seed(5)
+    dev = "cuda"
+    ext = hip_ext()
+    p32 = torch.randn(1000, device=dev)
+    g = torch.randn(1000, device=dev)
+    m = torch.zeros(1000, device=dev)
+    v = torch.zeros(1000, device=dev)
+    pb = p32.to(torch.bfloat16).clone()
+    master = pb.float().clone()
+    m2 = torch.zeros(1000, device=dev)
+    v2 = torch.zeros(1000, device=dev)
+    gb = g.to(torch.bfloat16)
+    lr, b1, b2, eps = 1e-2, 0.9, 0.95, 1e-8
+    # reference: fp32 path fed the SAME bf16-rounded inputs
+    pref = master.clone()
+    gref = gb.float()
+    ext.adamw_step([pref], [gref], [m, v][:1], [v], lr, b1, b2, eps, 0.0,
+                   1 - b1, 1 - b2, [])
+    ext.adamw_step([pb], [gb], [m2], [v2], lr, b1, b2, eps, 0.0,
+                   1 - b1, 1 - b2, [master])
+    assert torch.allclose(master, pref, atol=1e-6), (master - pref).abs().max()
+    assert torch.equal(pb, master.to(torch.bfloat16))
