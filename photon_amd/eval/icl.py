@@ -26,9 +26,20 @@ import torch
 import torch.nn.functional as F
 
 
+def _resolve_task_path(path: str | Path) -> Path:
+    """Relative task paths resolve against the photon_amd package dir
+    (bundled demo tasks), falling back to the cwd."""
+    p = Path(path)
+    if not p.is_absolute() and not p.exists():
+        pkg = Path(__file__).resolve().parent.parent / p
+        if pkg.exists():
+            return pkg
+    return p
+
+
 def load_jsonl_task(path: str | Path) -> list[dict]:
     out = []
-    with open(path) as f:
+    with open(_resolve_task_path(path)) as f:
         for line in f:
             line = line.strip()
             if line:
