@@ -49,10 +49,8 @@ def main(argv: list[str] | None = None):
         else torch.device("cpu")
     )
     server = FedServer(cfg, comm, device)
-    history = server.run()
+    server.run()
     if rank == 0:
-        last = max(history.losses_distributed, default=None) if hasattr(
-            history, "losses_distributed") else None
         print(f"[fed] finished {server.n_rounds} rounds "
               f"(run_uuid={server.run_uuid})")
     return server
