@@ -16,6 +16,7 @@ from ..models import build_model
 from ..train import Trainer
 from .flat import FlatParams
 from .params_ops import (
+    freeze_blocks,
     manipulate_pre_training,
     post_process_client_result,
     set_optimizer_state,
@@ -65,6 +66,11 @@ class FedClient:
     def _ensure_trainer(self, cid: int) -> Trainer:
         llm = self.cfg["llm_config"]
         if self.trainer is None:
+            fl = self.cfg.get("fl", {})
+            if fl.get("frozen_layers"):
+                # fl.frozen/unfrozen_layers (photon/utils.py:322-387)
+                freeze_blocks(self.model, fl.get("frozen_layers"),
+                              fl.get("unfrozen_layers"))
             self.trainer = Trainer(
                 self.model,
                 llm,
@@ -107,7 +113,7 @@ class FedClient:
         from ..train.timestamp import Timestamp
 
         trainer.timestamp = Timestamp()
-        if cid in self._timestamps:
+        if cid in self._timestamps and not fl.get("reset_timestamp", False):
             trainer.timestamp.load_state_dict(self._timestamps[cid])
         if cid in self._loader_states and not fl.get("reset_dataset_state", False):
             trainer.train_loader.load_state_dict(self._loader_states[cid])
@@ -228,4 +234,53 @@ class FedClient:
         metrics = trainer.eval(subset_num_batches)
         loss = metrics.get("metrics/eval/LanguageCrossEntropy", float("nan"))
         n = float(metrics.get("eval_samples", 0))
+        if self.cfg.get("fl", {}).get("use_unigram_metrics", False):
+            metrics.update(self._unigram_metrics(cid, trainer, loss,
+                                                 subset_num_batches))
         return loss, n, metrics
+
+    @torch.no_grad()
+    def _unigram_metrics(self, cid: int, trainer, model_ce: float,
+                         subset_num_batches: int) -> dict:
+        """Unigram-normalized eval CE (reference unigram metrics path,
+        SURVEY.md §5.5): model CE minus the unigram CE of the SAME labels,
+        using the client's 1_gram.json written at dataset conversion.
+        Failures are tolerated per fl.allow_unigram_metrics_failures."""
+        import math
+        from pathlib import Path
+
+        from ..metrics import (
+            PureUnigramCrossEntropy,
+            load_freq_map,
+            unigram_tensor_from_freq,
+        )
+
+        try:
+            split_cfg = self.cfg["dataset"]["train"]
+            root = Path(split_cfg.get("root_local") or "")
+            freq_path = root / f"client_{cid}" / "1_gram.json"
+            vocab = int(self.cfg["llm_config"]["model"].get("vocab_size", 50368))
+            probs = unigram_tensor_from_freq(load_freq_map(freq_path), vocab)
+            metric = PureUnigramCrossEntropy(probs)
+            # same window the eval loop used
+            loader = trainer.eval_loader
+            state = loader.state_dict()
+            nb = subset_num_batches if subset_num_batches > 0 else 8
+            loader.load_state_dict(
+                {"samples_consumed":
+                 max(0, state["samples_consumed"] - nb * loader.batch_size)}
+            )
+            for _ in range(nb):
+                metric.update(loader.next_batch()["input_ids"][:, 1:])
+            unigram_ce = metric.compute()
+            return {
+                "metrics/eval/PureUnigramCrossEntropy": unigram_ce,
+                "metrics/eval/UnigramNormalizedLanguageCrossEntropy":
+                    model_ce - unigram_ce,
+                "metrics/eval/UnigramNormalizedLanguagePerplexity":
+                    math.exp(model_ce - unigram_ce),
+            }
+        except (OSError, KeyError, ValueError) as e:
+            if self.cfg["fl"].get("allow_unigram_metrics_failures", True):
+                return {}
+            raise

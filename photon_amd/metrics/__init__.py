@@ -1,4 +1,5 @@
 from .unigram import (
+    load_freq_map,
     PureUnigramCrossEntropy,
     PureUnigramPerplexity,
     UnigramNormalizedLanguageCrossEntropy,
@@ -9,6 +10,7 @@ from .unigram import (
 )
 
 __all__ = [
+    "load_freq_map",
     "PureUnigramCrossEntropy",
     "PureUnigramPerplexity",
     "UnigramNormalizedLanguageCrossEntropy",

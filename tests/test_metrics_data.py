@@ -151,3 +151,51 @@ def test_icl_language_modeling_and_mc(tmp_path):
     ]}
     comp = gauntlet_composite(res, gauntlet)
     assert "metrics/eval_gauntlet/average" in comp
+
+
+def test_fed_unigram_metrics_end_to_end(tmp_path, tiny_llm_config=None):
+    """use_unigram_metrics through the fed evaluate path: convert a corpus
+    (writes 1_gram.json), train-eval a tiny model on the shards, and check
+    the unigram-normalized CE appears and is ~ model_CE - unigram_CE."""
+    import copy
+
+    from photon_amd.conf import compose, config_yaml_dir
+    from photon_amd.data.convert import convert
+    from photon_amd.fed.runtime import Comm
+    from photon_amd.fed.server import FedServer
+
+    convert("synthetic:64", tmp_path / "corpus", num_clients=2,
+            concat_tokens=64, split="train")
+    convert("synthetic:16", tmp_path / "corpus", num_clients=2,
+            concat_tokens=64, split="validation", seed=5)
+    cfg = compose(config_yaml_dir(), "base", ["llm_config=mpt-125m"]).to_plain()
+    llm = cfg["llm_config"]
+    llm["model"].update({"d_model": 64, "n_heads": 2, "n_layers": 2,
+                         "max_seq_len": 64, "vocab_size": 258})
+    llm["model"]["attn_config"]["attn_impl"] = "torch"
+    llm.update({"global_train_batch_size": 2, "device_train_microbatch_size": 2,
+                "local_steps": "1ba", "precision": "fp32",
+                "eval_subset_num_batches": 2, "device_eval_batch_size": 2})
+    streams = [
+        {"client_streams": {"stream_0": {"local": "client_0"}}},
+        {"client_streams": {"stream_1": {"local": "client_1"}}},
+    ]
+    cfg["dataset"] = {
+        "train": {"split": "train", "root_local": str(tmp_path / "corpus"),
+                  "streams": streams, "shuffle": False},
+        "val": {"split": "validation", "root_local": str(tmp_path / "corpus"),
+                "streams": streams, "shuffle": False},
+    }
+    cfg["fl"].update({"n_total_clients": 2, "n_clients_per_round": 2,
+                      "n_rounds": 1, "use_unigram_metrics": True,
+                      "allow_unigram_metrics_failures": False})
+    cfg["photon"]["checkpoint"] = False
+    cfg["photon"]["saving_path"] = str(tmp_path / "ck")
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    srv.run_round(1)
+    loss, n, metrics = srv.client.evaluate(0, srv.strategy.params, srv.layout, 2)
+    assert "metrics/eval/UnigramNormalizedLanguageCrossEntropy" in metrics
+    norm = metrics["metrics/eval/UnigramNormalizedLanguageCrossEntropy"]
+    uni = metrics["metrics/eval/PureUnigramCrossEntropy"]
+    assert abs((loss - uni) - norm) < 1e-9
