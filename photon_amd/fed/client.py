@@ -105,6 +105,10 @@ class FedClient:
         llm = self.cfg["llm_config"]
         fl = self.cfg.get("fl", {})
         momenta = bool(fl.get("aggregate_momenta", False))
+        # duration precedence: explicit arg > fl.n_local_steps > llm_config
+        # (reference n_local_steps/local_steps duality, base_schema.py FL)
+        if local_steps is None and int(fl.get("n_local_steps", 0) or 0) > 0:
+            local_steps = int(fl["n_local_steps"])
         steps = duration_to_batches(
             local_steps if local_steps is not None else llm.get("local_steps", "500ba")
         )
@@ -153,7 +157,8 @@ class FedClient:
 
         t0 = time.time()
         global_flat, m1_in, m2_in = manipulate_pre_training(
-            payload, layout, fl, cid, local_params=self._personal.get(cid)
+            payload, layout, fl, cid, local_params=self._personal.get(cid),
+            server_round=server_round,
         )
         # set params from the global buffer (HBM->HBM copies, no host hop);
         # with bf16 weights the fp32 optimizer masters get the EXACT global

@@ -203,6 +203,7 @@ def manipulate_pre_training(
     fl_cfg: dict,
     cid: int,
     local_params: torch.Tensor | None = None,
+    server_round: int | None = None,
 ):
     """Split + transform the incoming payload before local training.
 
@@ -213,8 +214,15 @@ def manipulate_pre_training(
     params, m1, m2 = split_payload(payload, layout.total, momenta)
     params = params.clone()
     rand = fl_cfg.get("random_layers")
-    if rand:
-        randomize_layers(layout, params, rand, seed=int(fl_cfg.get("seed", 0)) + cid)
+    freq = int(fl_cfg.get("random_init_freq", 0) or 0)
+    due = freq <= 0 or (server_round is not None and server_round % freq == 0)
+    if rand and due:
+        # truly_random_init: fresh randomness each round; else the same
+        # deterministic per-client init every time (reference flag pair)
+        seed = int(fl_cfg.get("seed", 0)) + cid
+        if fl_cfg.get("truly_random_init", True) and server_round is not None:
+            seed += server_round * 10007
+        randomize_layers(layout, params, rand, seed=seed)
     pers = fl_cfg.get("personalized_layers")
     if pers and local_params is not None:
         personalize_layers(layout, params, local_params, pers)
