@@ -13,6 +13,7 @@ import torch
 from ..conf.schema import duration_to_batches
 from ..data import build_eval_loader, build_train_loader
 from ..models import build_model
+from ..models.mpt import resize_vocab
 from ..train import Trainer
 from .flat import FlatParams
 from .params_ops import (
@@ -43,6 +44,9 @@ class FedClient:
         llm = cfg["llm_config"]
         torch.manual_seed(int(llm.get("seed", 17)))
         self.model = build_model(llm)
+        rv = int(cfg.get("fl", {}).get("resize_vocab") or 0)
+        if rv:
+            resize_vocab(self.model, rv)
         self.trainer: Trainer | None = None
         self.rank = rank
         from pathlib import Path

@@ -198,6 +198,24 @@ class MPTCausalLM(nn.Module):
         )
 
 
+@torch.no_grad()
+def resize_vocab(model: MPTCausalLM, new_vocab: int) -> MPTCausalLM:
+    """Resize the tied wte/LM-head vocabulary (reference fl.resize_vocab,
+    base_schema.py FL): existing rows kept, new rows init-normal. Changes
+    the wire format size — all participants must use the same value."""
+    old = model.transformer.wte
+    if new_vocab == old.num_embeddings:
+        return model
+    new = nn.Embedding(new_vocab, old.embedding_dim,
+                       device=old.weight.device, dtype=old.weight.dtype)
+    nn.init.normal_(new.weight, mean=0.0, std=model.cfg.init_std)
+    keep = min(new_vocab, old.num_embeddings)
+    new.weight.data[:keep] = old.weight.data[:keep]
+    model.transformer.wte = new
+    model.cfg.vocab_size = new_vocab
+    return model
+
+
 def build_model(llm_config: dict) -> MPTCausalLM:
     """Build from the llm_config subtree (photon_amd.conf)."""
     cfg = MPTConfig.from_cfg(llm_config["model"])

@@ -110,3 +110,19 @@ def test_tied_lm_head():
     logits = m(ids)["logits"]
     manual = m.transformer(ids) @ m.transformer.wte.weight.t()
     assert torch.allclose(logits, manual, atol=1e-5)
+
+
+def test_resize_vocab():
+    from photon_amd.models.mpt import MPTCausalLM, MPTConfig, resize_vocab
+
+    torch.manual_seed(0)
+    m = MPTCausalLM(MPTConfig(d_model=64, n_heads=2, n_layers=1,
+                              max_seq_len=32, vocab_size=100,
+                              attn_impl="torch", loss_impl="torch"))
+    old_rows = m.transformer.wte.weight[:100].clone()
+    resize_vocab(m, 140)
+    assert m.transformer.wte.num_embeddings == 140
+    assert torch.equal(m.transformer.wte.weight[:100], old_rows)
+    ids = torch.randint(0, 140, (2, 16))
+    out = m(ids, labels=ids)
+    assert torch.isfinite(out["loss"])
