@@ -98,3 +98,13 @@ def tiny_cfg(tiny_llm_config, tmp_path):
 @pytest.fixture(autouse=True)
 def _deterministic():
     torch.manual_seed(0)
+
+
+def free_port() -> int:
+    """Pick an OS-assigned free TCP port (avoids fixed-port flakes when the
+    suite runs back-to-back or alongside other jobs)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]

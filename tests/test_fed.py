@@ -143,7 +143,8 @@ def test_two_rank_gloo_equivalence(tiny_cfg, tmp_path):
     ctx = mp.get_context("spawn")
     out_dir = str(tmp_path / "out")
     os.makedirs(out_dir, exist_ok=True)
-    port = 29511
+    from tests.conftest import free_port
+    port = free_port()
     procs = [
         ctx.Process(target=_dist_worker, args=(r, 2, port, cfg2, out_dir))
         for r in range(2)

@@ -78,7 +78,8 @@ def test_tp_two_rank_matches_unsharded(tmp_path):
     ref_wte_grad = ref.transformer.wte.weight.grad.clone()
 
     ctx = mp.get_context("spawn")
-    port = 29533
+    from tests.conftest import free_port
+    port = free_port()
     procs = [
         ctx.Process(target=_tp_worker, args=(r, 2, port, str(tmp_path)))
         for r in range(2)
@@ -123,7 +124,8 @@ def _ddp_worker(rank, world, port, out_dir):
 @pytest.mark.timeout(600)
 def test_ddp_two_rank_sync(tmp_path):
     ctx = mp.get_context("spawn")
-    port = 29534
+    from tests.conftest import free_port
+    port = free_port()
     procs = [
         ctx.Process(target=_ddp_worker, args=(r, 2, port, str(tmp_path)))
         for r in range(2)
