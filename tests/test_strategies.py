@@ -131,3 +131,72 @@ def test_metrics_present(layout):
     assert "l2_norm_pseudo_gradient" in m
     assert "l2_norm_momentum_vector" in m
     assert len(m["layerwise_l2_norms_pseudo_gradient"]) == len(layout.names)
+
+
+# ---------------------------------------------------------------------------
+# Property-based fuzzing (hypothesis): the strategies must match their NumPy
+# oracles for arbitrary hyperparameters and multi-round sequences.
+# ---------------------------------------------------------------------------
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(
+    slr=st.floats(0.05, 1.5),
+    mu=st.floats(0.0, 0.99),
+    rounds=st.integers(1, 4),
+    seed=st.integers(0, 10_000),
+)
+def test_nesterov_property(layout, slr, mu, rounds, seed):
+    rng = np.random.default_rng(seed)
+    x0 = rng.standard_normal(layout.total).astype(np.float32)
+    strat = dispatch_strategy("NESTOROV", layout,
+                              {"server_learning_rate": slr, "server_momentum": mu})
+    strat.initialize(torch.from_numpy(x0.copy()))
+    x = x0.copy()
+    m = np.zeros_like(x)
+    for r in range(1, rounds + 1):
+        avg = rng.standard_normal(layout.total).astype(np.float32)
+        strat.update(torch.from_numpy(avg.copy()), r, 2)
+        g = x - avg
+        m = mu * m + g
+        gp = g + mu * m
+        x = x - slr * gp
+    assert np.allclose(strat.params.numpy(), x, atol=1e-4), (
+        np.abs(strat.params.numpy() - x).max()
+    )
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(
+    eta=st.floats(1e-3, 0.5),
+    b1=st.floats(0.0, 0.95),
+    b2=st.floats(0.5, 0.999),
+    tau=st.floats(1e-6, 1e-2),
+    rounds=st.integers(1, 4),
+    seed=st.integers(0, 10_000),
+)
+def test_fedadam_property(layout, eta, b1, b2, tau, rounds, seed):
+    rng = np.random.default_rng(seed)
+    x0 = rng.standard_normal(layout.total).astype(np.float32)
+    strat = dispatch_strategy("FEDADAM", layout, {
+        "eta": eta, "beta_1": b1, "beta_2": b2, "tau": tau})
+    strat.initialize(torch.from_numpy(x0.copy()))
+    x = x0.copy()
+    m = np.zeros_like(x)
+    v = np.zeros_like(x)
+    for r in range(1, rounds + 1):
+        avg = rng.standard_normal(layout.total).astype(np.float32)
+        strat.update(torch.from_numpy(avg.copy()), r, 2)
+        g = x - avg
+        m = b1 * m + (1 - b1) * g
+        v = b2 * v + (1 - b2) * g * g
+        mh = m / (1 - b1**r)
+        vh = v / (1 - b2**r)
+        x = x - eta * mh / (np.sqrt(vh) + tau)
+    assert np.allclose(strat.params.numpy(), x, atol=1e-4), (
+        np.abs(strat.params.numpy() - x).max()
+    )
