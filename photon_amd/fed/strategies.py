@@ -10,8 +10,10 @@ oracles in tests/test_strategies.py):
                                      x <- x + eta * m_hat / (sqrt(v_hat) + tau)
                                      (sic: the reference ADDS along the
                                      x-minus-avg pseudo-gradient — verified
-                                     against fedadam.py:291-316; parity
-                                     preserves the exact rule)
+                                     against fedadam.py:291-316, and FedYogi
+                                     does the same at fedyogi.py:297-316;
+                                     parity preserves the exact rules —
+                                     NESTOROV, the default, subtracts)
 * FEDYOGI  (fedyogi.py:299-320):     v += (1-b2)*g^2*sign(g^2 - v); x as FedAdam
 
 where g = x - sf*avg, sf = scaling_fn(n_clients) in {1, linear, sqrt}
