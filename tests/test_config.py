@@ -86,3 +86,18 @@ def test_resolver_cli(tmp_path, monkeypatch):
     out = main(["fl.n_rounds=3"])
     cfg = load_resolved(out)
     assert cfg.fl.n_rounds == 3
+
+
+# -- property fuzzing --------------------------------------------------------
+from hypothesis import given, settings
+from hypothesis import strategies as hst
+
+from photon_amd.conf.schema import duration_to_batches
+
+
+@settings(max_examples=30, deadline=None)
+@given(n=hst.integers(0, 10**7))
+def test_duration_parse_property(n):
+    assert duration_to_batches(f"{n}ba") == n
+    assert duration_to_batches(n) == n
+    assert duration_to_batches(str(n)) == n

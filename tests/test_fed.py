@@ -267,3 +267,25 @@ def test_resume_restores_client_state(tiny_cfg, tmp_path):
     srv2.initialize()
     assert {c: s.steps_done for c, s in srv2.client.client_states.items()} == steps
     assert srv2.server_steps_cumulative == srv.server_steps_cumulative
+
+
+def test_npz_roundtrip_property(tiny_cfg):
+    """FlatParams .npz save/load round-trips bit-exactly for several
+    random payloads (the server-checkpoint wire format)."""
+    import numpy as np
+    import tempfile
+
+    from photon_amd.models import build_model
+
+    model = build_model(tiny_cfg["llm_config"])
+    layout = FlatParams(model)
+    for seed in range(3):
+        torch.manual_seed(seed)
+        flat = torch.randn_like(layout.flat)
+        with tempfile.TemporaryDirectory() as d:
+            p = os.path.join(d, "x.npz")
+            layout.save_npz(p, flat)
+            arrays = layout.load_npz(p)
+            back = torch.cat([torch.from_numpy(np.ascontiguousarray(a)).reshape(-1)
+                              for a in arrays])
+            assert torch.equal(back, flat)
