@@ -82,17 +82,29 @@ class TokenShardDataset:
         ]
         self._offsets = np.cumsum([0] + [s["num_tokens"] for s in self.index["shards"]])
         self.num_samples = self.index["num_tokens"] // seq_len
-        if shuffle:
-            rng = np.random.default_rng(shuffle_seed)
-            self.order = rng.permutation(self.num_samples)
-        else:
-            self.order = np.arange(self.num_samples)
+        self.shuffle = shuffle
+        self.shuffle_seed = shuffle_seed
+        self._order_epoch = -1
+        self._order = np.arange(self.num_samples)
+
+    def _order_for(self, epoch: int) -> np.ndarray:
+        # Deterministic PER-EPOCH reshuffle (mosaicml-streaming's
+        # epoch-seeded shuffle semantics): order = perm(seed + epoch),
+        # reproducible from any resume point.
+        if not self.shuffle:
+            return self._order
+        if epoch != self._order_epoch:
+            rng = np.random.default_rng(self.shuffle_seed + epoch)
+            self._order = rng.permutation(self.num_samples)
+            self._order_epoch = epoch
+        return self._order
 
     def __len__(self) -> int:
         return self.num_samples
 
     def __getitem__(self, i: int) -> torch.Tensor:
-        sample = int(self.order[i % self.num_samples])
+        epoch = int(i) // self.num_samples
+        sample = int(self._order_for(epoch)[int(i) % self.num_samples])
         start = sample * self.seq_len
         end = start + self.seq_len
         # locate shard(s); windows can straddle shard boundaries

@@ -105,3 +105,35 @@ def test_convert_then_train_from_shards(tmp_path, tiny_llm_config):
     loader1 = build_train_loader(cfg, client_id=1, batch_size=2)
     b1 = loader1.next_batch()
     assert not torch.equal(batch["input_ids"], b1["input_ids"])
+
+
+def test_shard_epoch_reshuffle(tmp_path):
+    """Shuffled shard datasets reshuffle deterministically per epoch and
+    reproduce the same order from a resumed position."""
+    from photon_amd.data.convert import convert
+    from photon_amd.data.shards import StatefulLoader, TokenShardDataset
+
+    convert("synthetic:32", tmp_path, num_clients=1, concat_tokens=64)
+    d = tmp_path / "client_0" / "train"
+    ds = TokenShardDataset(d, seq_len=64, shuffle=True, shuffle_seed=3)
+    n = len(ds)
+    assert n >= 4
+    epoch0 = [ds[i] for i in range(n)]
+    epoch1 = [ds[n + i] for i in range(n)]
+    # different order across epochs (same multiset of samples)
+    assert any(not torch.equal(a, b) for a, b in zip(epoch0, epoch1))
+    s0 = {bytes(t.numpy().tobytes()) for t in epoch0}
+    s1 = {bytes(t.numpy().tobytes()) for t in epoch1}
+    assert s0 == s1
+    # resume reproducibility
+    ds2 = TokenShardDataset(d, seq_len=64, shuffle=True, shuffle_seed=3)
+    assert torch.equal(ds2[n + 2], epoch1[2])
+    # loader state round-trips across the epoch boundary
+    loader = StatefulLoader(ds, 2)
+    for _ in range(n // 2):
+        loader.next_batch()
+    state = loader.state_dict()
+    b_next = loader.next_batch()
+    loader2 = StatefulLoader(ds2, 2)
+    loader2.load_state_dict(state)
+    assert torch.equal(loader2.next_batch()["input_ids"], b_next["input_ids"])
