@@ -107,6 +107,9 @@ def main(argv: list[str] | None = None):
     layout.copy_from_model(model)
     comm.broadcast_flat(layout.flat, src=0)
     layout.copy_to_model(model)
+    if cent.get("store_init_model", False) and rank == 0:
+        (save_path / run_uuid).mkdir(parents=True, exist_ok=True)
+        layout.save_npz(save_path / run_uuid / "init_parameters.npz")
 
     # pretrained init: Composer .pt checkpoint, or .npz parameter dump
     # (reference centralised_train.py:98-117, incl. the WTE-only transplant).
