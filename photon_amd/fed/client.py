@@ -131,7 +131,10 @@ class FedClient:
         # (reference llm_config_functions.py:642-764, clients/utils.py:217-228).
         ckpt_dir = self.save_root / f"client_{cid}"
         expect = trainer.timestamp.batch + steps
-        skip_path = ckpt_dir / f"ep0-ba{expect}-rank{self.rank}.pt"
+        # epoch-aware probe (not hard-coded ep0): the restored per-client
+        # timestamp carries the epoch the checkpoint was written at
+        ep = trainer.timestamp.epoch
+        skip_path = ckpt_dir / f"ep{ep}-ba{expect}-rank{self.rank}.pt"
         if self.save_client_checkpoints and skip_path.exists():
             trainer.save_folder = ckpt_dir
             trainer.load_checkpoint(
@@ -164,6 +167,11 @@ class FedClient:
             payload, layout, fl, cid, local_params=self._personal.get(cid),
             server_round=server_round,
         )
+        # Optimizer reset FIRST: state.clear() would otherwise wipe the
+        # exact-fp32 masters that sync_masters injects below (the round's
+        # parameter set must not round-trip through bf16).
+        if reset_optimizer:
+            trainer.optimizer.state.clear()
         # set params from the global buffer (HBM->HBM copies, no host hop);
         # with bf16 weights the fp32 optimizer masters get the EXACT global
         # values (no bf16 round-trip)
@@ -178,8 +186,6 @@ class FedClient:
                     params[n].data.copy_(v.to(params[n].dtype))
         set_params_time = time.time() - t0
 
-        if reset_optimizer:
-            trainer.optimizer.state.clear()
         if momenta and m1_in is not None:
             # import aggregated momenta + step for bias correction
             st_prev = self.client_states.get(cid)

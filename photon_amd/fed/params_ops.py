@@ -134,31 +134,48 @@ def get_optimizer_momenta(
 # ---------------------------------------------------------------------------
 # Layer selection: freeze / personalize / randomize
 # ---------------------------------------------------------------------------
+def _pattern_matches(name: str, p: str) -> bool:
+    """Anchored pattern match: exact name, glob, or ``.``-boundary
+    substring. Boundary anchoring means 'blocks.1' matches
+    'transformer.blocks.1.attn.Wqkv.weight' but NOT 'blocks.10.…', and a
+    bare index like '1' only matches a whole dotted component."""
+    if p == name:
+        return True
+    if any(ch in p for ch in "*?["):
+        return fnmatch.fnmatch(name, p)
+    start = 0
+    while True:
+        i = name.find(p, start)
+        if i < 0:
+            return False
+        before_ok = i == 0 or name[i - 1] == "."
+        j = i + len(p)
+        after_ok = j == len(name) or name[j] == "."
+        if before_ok and after_ok:
+            return True
+        start = i + 1
+
+
 def _match_names(names: list[str], patterns) -> list[str]:
-    """Names matching any glob/substring pattern (reference passes either
-    exact names, substrings, or block indices)."""
-    out = []
-    for n in names:
-        for pat in patterns or []:
-            p = str(pat)
-            if p == n or p in n or fnmatch.fnmatch(n, p):
-                out.append(n)
-                break
-    return out
+    """Names matching any pattern (reference passes exact names,
+    substrings, or block indices — anchored here, see _pattern_matches)."""
+    pats = [str(p) for p in (patterns or [])]
+    return [n for n in names if any(_pattern_matches(n, p) for p in pats)]
 
 
 def freeze_blocks(model: torch.nn.Module, frozen, unfrozen=None) -> list[str]:
-    """Set requires_grad by name patterns (photon/utils.py:322-387):
-    ``frozen`` patterns freeze, ``unfrozen`` re-enables (takes precedence)."""
+    """Set requires_grad by name patterns with the reference's semantics
+    (photon/utils.py:368-387): a param is frozen when it matches ``frozen``
+    OR when ``unfrozen`` is given and it does NOT match ``unfrozen``
+    (i.e. providing unfrozen_layers freezes the complement)."""
     names = [n for n, _ in model.named_parameters()]
-    to_freeze = set(_match_names(names, frozen))
-    to_unfreeze = set(_match_names(names, unfrozen)) if unfrozen else set()
+    to_freeze = set(_match_names(names, frozen)) if frozen else set()
+    keep = set(_match_names(names, unfrozen)) if unfrozen else None
     touched = []
     for n, p in model.named_parameters():
-        if n in to_unfreeze:
-            p.requires_grad_(True)
-            touched.append(n)
-        elif n in to_freeze:
+        if not p.requires_grad:
+            continue
+        if n in to_freeze or (keep is not None and n not in keep):
             p.requires_grad_(False)
             touched.append(n)
     return touched

@@ -87,6 +87,15 @@ class Comm:
         dist.all_gather(out, t)
         return [float(x.item()) for x in out]
 
+    def all_gather_obj(self, obj):
+        """All-gather an arbitrary picklable object (per-cid eval rows);
+        returns a list of world_size objects."""
+        if not self.is_distributed:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj)
+        return out
+
     def all_reduce_(self, flat: torch.Tensor) -> torch.Tensor:
         if self.is_distributed:
             dist.all_reduce(flat, op=dist.ReduceOp.SUM)
@@ -109,6 +118,12 @@ class Comm:
         return local_weighted_sum, total
 
 
+# Incremental sampler cache: one Random stream per (seed, n_total,
+# n_per_round), advanced round by round — O(1) per round instead of
+# replaying from round 1 on every call (O(n_rounds^2) total).
+_SAMPLER_CACHE: dict[tuple, tuple[random.Random, int, list[int]]] = {}
+
+
 def sample_clients(
     seed: int, current_round: int, n_total: int, n_per_round: int
 ) -> list[int]:
@@ -116,13 +131,19 @@ def sample_clients(
 
     Mirrors the reference's server-side PRNG fast-forward semantics
     (photon/server_app.py:188-192,295): one Random(seed) stream advanced
-    round by round so resume reproduces the same schedule.
+    round by round so resume reproduces the same schedule. The stream is
+    cached and advanced incrementally; asking for an earlier round than
+    the cache has reached replays from scratch (resume path, happens once).
     """
-    rng = random.Random(seed)
-    sampled: list[int] = []
-    for _ in range(current_round):
-        sampled = rng.sample(range(n_total), n_per_round)
-    return sorted(sampled)
+    key = (seed, n_total, n_per_round)
+    rng, done, last = _SAMPLER_CACHE.get(key, (None, 0, []))
+    if rng is None or current_round < done:
+        rng, done, last = random.Random(seed), 0, []
+    while done < current_round:
+        last = rng.sample(range(n_total), n_per_round)
+        done += 1
+    _SAMPLER_CACHE[key] = (rng, done, last)
+    return sorted(last)
 
 
 def assign_clients_to_ranks(sampled: list[int], world_size: int) -> dict[int, list[int]]:

@@ -43,6 +43,7 @@ from .server_ckpt import (
     copy_old_checkpoints_to_new_run,
     delete_rounds,
     interpret_resume_round,
+    load_client_momenta,
     obtain_sorted_rounds,
     resume_from_round,
     upload_server_checkpoint,
@@ -159,6 +160,16 @@ class FedServer:
                         )
                 except (ValueError, SyntaxError):
                     pass
+            # restore aggregated client momenta (fl.aggregate_momenta):
+            # without this the first post-resume round would broadcast zero
+            # momenta straight into the clients' Adam state
+            if self.aggregate_momenta:
+                cm = load_client_momenta(
+                    self.saving_path, self.run_uuid, target, self.layout
+                )
+                if cm is not None:
+                    self.client_m1.copy_(cm[0])
+                    self.client_m2.copy_(cm[1])
             self.start_round = target + 1
             resumed = target
         else:
@@ -288,34 +299,60 @@ class FedServer:
         return round_metrics
 
     def evaluate_round(self, server_round: int) -> float:
-        """Reference evaluate_round (evaluate_utils.py:232): every rank
-        evaluates its client shard; weighted_loss_avg across ranks. With
-        fl.split_eval each client's own stream is also reported per-cid
-        (the reference's split evaluation)."""
+        """Reference evaluate_round (evaluate_utils.py:232 +
+        node_manager_app.py:594-725): EVERY sampled client is evaluated —
+        each rank loops over its round-robin shard of the sampled cids,
+        per-cid (loss, n, metrics) rows are all-gathered and aggregated with
+        weighted_loss_avg. ``eval_subset_num_batches: -1`` means the full
+        eval split (trainer semantics), not a coerced window."""
         subset = int(self.cfg["llm_config"].get("eval_subset_num_batches", -1))
-        if subset <= 0:
-            subset = 8
-        cid = self.comm.rank if self.comm.rank < self.n_total else 0
-        loss, n, metrics = self.client.evaluate(
-            cid, self.strategy.params, self.layout, subset
+        sampled = sample_clients(
+            self.seed, server_round, self.n_total, self.n_per_round
         )
-        losses = self.comm.all_gather_scalars(loss)
-        weights = self.comm.all_gather_scalars(n)
-        avg = weighted_loss_avg(list(zip(losses, weights)))
+        assignment = assign_clients_to_ranks(sampled, self.comm.world_size)
+        local_rows = []
+        for cid in assignment[self.comm.rank]:
+            loss, n, metrics = self.client.evaluate(
+                cid, self.strategy.params, self.layout, subset
+            )
+            local_rows.append(
+                (cid, float(loss), float(n),
+                 {k: v for k, v in metrics.items()
+                  if isinstance(v, (int, float))})
+            )
+        rows = [r for per_rank in self.comm.all_gather_obj(local_rows)
+                for r in per_rank]
+        avg = weighted_loss_avg([(l, n) for _, l, n, _ in rows])
         if self.comm.rank == 0:
             self.history.add_loss_distributed(server_round, avg)
             extra = {}
             if self.cfg["fl"].get("split_eval", False):
                 extra = {
-                    f"metrics/eval/LanguageCrossEntropy_client_{i}": l
-                    for i, l in enumerate(losses)
+                    f"metrics/eval/LanguageCrossEntropy_client_{cid}": l
+                    for cid, l, _, _ in rows
                 }
-            # config-gated ICL / gauntlet evaluation on the global model
+            # unigram metric aggregation across clients (the gather above is
+            # the RCCL analogue of torchmetrics dist_reduce_fx="sum")
+            for key in (
+                "metrics/eval/PureUnigramCrossEntropy",
+                "metrics/eval/UnigramNormalizedLanguageCrossEntropy",
+                "metrics/eval/UnigramNormalizedLanguagePerplexity",
+            ):
+                vals = [(m[key], n) for _, _, n, m in rows if key in m]
+                if vals:
+                    extra[key] = weighted_loss_avg(vals)
+            # config-gated ICL / gauntlet evaluation — explicitly on the
+            # CURRENT global params (strategy.params), never layout.flat
+            # which is only written at fresh init
             icl_cfg = self.cfg.get("icl_tasks_config") or {}
             if icl_cfg.get("icl_tasks"):
                 from ..centralised_train import run_icl_eval
 
-                self.layout.copy_to_model(self.client.model)
+                views = self.layout.layer_views_of(self.strategy.params)
+                params = dict(self.client.model.named_parameters())
+                with torch.no_grad():
+                    for n_, v in zip(self.layout.names, views):
+                        params[n_].data.copy_(v.to(params[n_].dtype))
                 extra.update(
                     run_icl_eval(self.cfg, self.client.model, self.device)
                 )
@@ -348,6 +385,11 @@ class FedServer:
                     self.history.state(),
                     {cid: vars(st) for cid, st in self.client.client_states.items()},
                     self.server_steps_cumulative,
+                    client_momenta=(
+                        (self.client_m1, self.client_m2)
+                        if self.aggregate_momenta
+                        else None
+                    ),
                 )
                 # per-round retention (reference cleanup_checkpoints_per_round,
                 # server_app.py:403-405): keep the newest N complete rounds

@@ -33,6 +33,16 @@ _KEY_FILES = {
     STATE_M2: "current_second_momentum_vector.npz",
 }
 
+# Aggregated CLIENT momenta (fl.aggregate_momenta): in the reference these
+# ride inside the parameters ndarray list and are therefore saved in
+# current_server_parameters.npz; the flat-buffer design splits them out into
+# their own files. Optional — not part of state_keys, so old checkpoints
+# stay resumable.
+_CLIENT_MOMENTA_FILES = (
+    "aggregated_client_momentum_vector.npz",
+    "aggregated_client_second_momentum_vector.npz",
+)
+
 
 def server_dir(saving_path: str | Path, run_uuid: str) -> Path:
     return Path(saving_path) / str(run_uuid) / "server"
@@ -48,6 +58,7 @@ def upload_server_checkpoint(
     client_state: dict,
     server_steps_cumulative: int,
     time_offset: float = 0.0,
+    client_momenta: tuple[torch.Tensor, torch.Tensor] | None = None,
 ) -> Path:
     rd = server_dir(saving_path, run_uuid) / str(server_round)
     rd.mkdir(parents=True, exist_ok=True)
@@ -62,7 +73,31 @@ def upload_server_checkpoint(
         pickle.dump(state, f)
     for key, tensor in strategy.state_tensors().items():
         layout.save_npz(rd / _KEY_FILES[key], tensor)
+    if client_momenta is not None:
+        for fname, tensor in zip(_CLIENT_MOMENTA_FILES, client_momenta):
+            layout.save_npz(rd / fname, tensor)
     return rd
+
+
+def load_client_momenta(
+    saving_path, run_uuid: str, server_round: int, layout: FlatParams,
+) -> tuple[torch.Tensor, torch.Tensor] | None:
+    """Restore aggregated client momenta saved by upload_server_checkpoint;
+    None when the round predates aggregate_momenta persistence."""
+    rd = server_dir(saving_path, run_uuid) / str(server_round)
+    paths = [rd / f for f in _CLIENT_MOMENTA_FILES]
+    if not all(p.exists() for p in paths):
+        return None
+    out = []
+    for p in paths:
+        arrays = layout.load_npz(p)
+        out.append(
+            torch.cat(
+                [torch.from_numpy(np.ascontiguousarray(a, dtype=np.float32)).reshape(-1)
+                 for a in arrays]
+            ).to(layout.flat.device)
+        )
+    return out[0], out[1]
 
 
 def obtain_sorted_rounds(saving_path, run_uuid: str, state_keys) -> list[int]:

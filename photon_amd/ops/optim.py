@@ -39,8 +39,10 @@ class _FusedStepMixin:
             if p.grad is None:
                 continue
             state = self.state[p]
-            if len(state) == 0:
-                state["step"] = 0
+            # NOT `len(state)==0`: sync_masters may have seeded only
+            # {'master': ...} before the first step.
+            if "exp_avg" not in state:
+                state.setdefault("step", 0)
                 state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                 state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
             if p.dtype != torch.float32 and "master" not in state:

@@ -222,16 +222,18 @@ class Trainer:
                 n = min(n, 1000)
             except TypeError:
                 n = 8
-        total, count = 0.0, 0
+        total_t, count = None, 0
         for _ in range(n):
             batch = self.eval_loader.next_batch()
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             with self.autocast():
                 out = self.model(ids, labels=ids)
-            total += float(out["loss"])
+            # accumulate on-device: ONE host sync after the loop, not per batch
+            l = out["loss"].detach()
+            total_t = l if total_t is None else total_t + l
             count += 1
         self.model.train()
-        loss = total / max(count, 1)
+        loss = float(total_t) / max(count, 1) if total_t is not None else 0.0
         metrics = {
             "metrics/eval/LanguageCrossEntropy": loss,
             "metrics/eval/LanguagePerplexity": float(torch.exp(torch.tensor(loss))),

@@ -289,3 +289,109 @@ def test_npz_roundtrip_property(tiny_cfg):
             back = torch.cat([torch.from_numpy(np.ascontiguousarray(a)).reshape(-1)
                               for a in arrays])
             assert torch.equal(back, flat)
+
+
+def test_eval_covers_every_sampled_client(tiny_cfg):
+    """VERDICT missing #5: on 1 rank with n sampled clients, evaluate_round
+    must evaluate EVERY sampled client (reference node_manager_app.py:594-725),
+    and split_eval reports a per-cid loss keyed by the real cid."""
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["fl"]["n_total_clients"] = 4
+    cfg["fl"]["n_clients_per_round"] = 3
+    cfg["fl"]["split_eval"] = True
+    cfg["photon"]["checkpoint"] = False
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    evaluated = []
+    real_eval = srv.client.evaluate
+
+    def spy(cid, *a, **k):
+        evaluated.append(cid)
+        return real_eval(cid, *a, **k)
+
+    srv.client.evaluate = spy
+    srv.run_round(1)
+    srv.evaluate_round(1)
+    sampled = sample_clients(cfg["seed"], 1, 4, 3)
+    assert sorted(evaluated) == sampled
+    # split_eval reported a per-cid metric for each sampled cid
+    per_cid = {
+        k for k in srv.history.metrics_distributed
+        if k.startswith("metrics/eval/LanguageCrossEntropy_client_")
+    }
+    assert per_cid == {
+        f"metrics/eval/LanguageCrossEntropy_client_{c}" for c in sampled
+    }
+
+
+def test_eval_full_split_not_coerced(tiny_cfg):
+    """eval_subset_num_batches=-1 must reach the trainer as -1 (full split),
+    not be coerced to 8 in the fed path (VERDICT weak #4)."""
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["llm_config"]["eval_subset_num_batches"] = -1
+    cfg["photon"]["checkpoint"] = False
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    seen = []
+    real_eval = srv.client.evaluate
+
+    def spy(cid, flat, layout, subset):
+        seen.append(subset)
+        return real_eval(cid, flat, layout, subset)
+
+    srv.client.evaluate = spy
+    srv.run_round(1)
+    srv.evaluate_round(1)
+    assert seen and all(s == -1 for s in seen)
+
+
+def test_aggregate_momenta_survive_resume(tiny_cfg, tmp_path):
+    """ADVICE m3: aggregated client momenta are persisted with the server
+    round checkpoint and restored on resume (no silent zero-momenta
+    broadcast after resume)."""
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["fl"]["aggregate_momenta"] = True
+    cfg["fl"]["reset_optimizer"] = False
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.run(2)
+    assert float(srv.client_m1.abs().sum()) > 0, "momenta should be non-zero"
+    m1, m2 = srv.client_m1.clone(), srv.client_m2.clone()
+
+    cfg2 = copy.deepcopy(cfg)
+    cfg2["photon"]["resume_round"] = -1
+    srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
+    srv2.initialize()
+    assert torch.equal(srv2.client_m1, m1)
+    assert torch.equal(srv2.client_m2, m2)
+
+
+def test_icl_eval_runs_on_current_global_params(tiny_cfg):
+    """ADVICE h1: the ICL/gauntlet eval must see the CURRENT global params
+    (strategy.params), not the stale layout.flat init snapshot."""
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = False
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    srv.run_round(1)
+    # simulate the ICL path's model refresh: patch run_icl_eval to capture
+    # the model's params at call time
+    captured = {}
+
+    def fake_icl(cfg_, model, device):
+        captured["flat"] = FlatParams(model).copy_from_model(model).flat.clone()
+        return {"icl/fake": 1.0}
+
+    import photon_amd.centralised_train as ct
+
+    orig = ct.run_icl_eval
+    ct.run_icl_eval = fake_icl
+    try:
+        srv.cfg["icl_tasks_config"] = {"icl_tasks": [{"label": "fake"}]}
+        srv.evaluate_round(1)
+    finally:
+        ct.run_icl_eval = orig
+        srv.cfg.pop("icl_tasks_config", None)
+    assert "flat" in captured
+    assert torch.allclose(captured["flat"], srv.strategy.params, atol=1e-6)

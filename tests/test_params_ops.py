@@ -105,16 +105,45 @@ def test_personalize_and_randomize_layers():
 
 def test_freeze_blocks():
     model = tiny_model()
-    touched = freeze_blocks(model, frozen=["blocks.0"], unfrozen=["blocks.0.norm_1"])
+    # frozen only: matching params freeze, everything else stays trainable
+    touched = freeze_blocks(model, frozen=["blocks.0"])
     assert touched
     for n, p in model.named_parameters():
-        if "blocks.0.norm_1" in n:
-            assert p.requires_grad
-        elif "blocks.0" in n:
-            assert not p.requires_grad
-    # restore
+        if ".blocks.0." in n:
+            assert not p.requires_grad, n
+        else:
+            assert p.requires_grad, n
     for p in model.parameters():
         p.requires_grad_(True)
+    # unfrozen given: the COMPLEMENT freezes (reference photon/utils.py:368-387
+    # — "name not in unfrozen_layers" freezes)
+    freeze_blocks(model, frozen=None, unfrozen=["blocks.0.norm_1"])
+    for n, p in model.named_parameters():
+        if "blocks.0.norm_1" in n:
+            assert p.requires_grad, n
+        else:
+            assert not p.requires_grad, n
+    for p in model.parameters():
+        p.requires_grad_(True)
+
+
+def test_match_names_boundary_anchoring():
+    from photon_amd.fed.params_ops import _match_names
+
+    names = [
+        "transformer.blocks.1.attn.Wqkv.weight",
+        "transformer.blocks.10.attn.Wqkv.weight",
+        "transformer.blocks.12.norm_1.weight",
+    ]
+    # 'blocks.1' must not swallow blocks.10/12
+    assert _match_names(names, ["blocks.1"]) == [names[0]]
+    # a bare index only matches a whole dotted component
+    assert _match_names(names, ["1"]) == [names[0]]
+    assert _match_names(names, ["10"]) == [names[1]]
+    # globs still work
+    assert _match_names(names, ["*norm_1*"]) == [names[2]]
+    # exact names
+    assert _match_names(names, [names[1]]) == [names[1]]
 
 
 def test_parameters_checker():

@@ -219,3 +219,24 @@ def test_microbatch_auto(tiny_llm_config, tmp_path):
     cfg["device_train_microbatch_size"] = "auto"
     tr = make_trainer(cfg, tmp_path)
     assert isinstance(tr.microbatch, int) and tr.microbatch >= 1
+
+
+def test_optimizer_bucket_with_master_only_state():
+    """ADVICE m2: after reset_optimizer clears state and sync_masters
+    re-seeds only {'master': ...}, the first step() must initialize
+    exp_avg/exp_avg_sq instead of crashing with KeyError."""
+    import torch
+
+    from photon_amd.ops.optim import DecoupledAdamW
+
+    p = torch.nn.Parameter(torch.randn(8))
+    opt = DecoupledAdamW([p], lr=1e-3)
+    # one normal step to build state, then a fed-style reset + master seed
+    p.grad = torch.randn(8)
+    opt.step()
+    opt.state.clear()
+    opt.sync_masters([p], [p.detach().to(torch.float32).clone()])
+    p.grad = torch.randn(8)
+    opt.step()  # must not raise
+    st = opt.state[p]
+    assert "exp_avg" in st and st["step"] == 1
