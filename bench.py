@@ -56,13 +56,42 @@ def main() -> None:
 
     from photon_amd.fed.rccl_tuning import apply_rccl_env
 
-    apply_rccl_env(cfg=None)
+    rccl_env = apply_rccl_env(cfg=None)
+
+    # ---- rendezvous / environment preflight (fail fast, loudly) ----------
+    import sys
+
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    env_world = int(os.environ.get("WORLD_SIZE", 1))
+    problems = []
+    if torch.cuda.is_available():
+        n_dev = torch.cuda.device_count()
+        if local_rank >= n_dev:
+            problems.append(f"LOCAL_RANK {local_rank} >= visible GPUs {n_dev}")
+        if env_world > 1 and os.environ.get("HSA_ENABLE_IPC_MODE_LEGACY") != "0":
+            # dmabuf IPC is required on this host driver; legacy IPC fails
+            # with hipIpcGetMemHandle: invalid argument
+            os.environ["HSA_ENABLE_IPC_MODE_LEGACY"] = "0"
+            print("[preflight] forced HSA_ENABLE_IPC_MODE_LEGACY=0",
+                  file=sys.stderr, flush=True)
+    if env_world > 1:
+        addr = os.environ.get("MASTER_ADDR", "")
+        if addr not in ("127.0.0.1", "localhost") and env_world > 1:
+            print(f"[preflight] MASTER_ADDR={addr!r} (expect 127.0.0.1 "
+                  "single-node)", file=sys.stderr, flush=True)
+    if problems:
+        raise SystemExit("[preflight] " + "; ".join(problems))
+
     rank, world = init_distributed()
     if world == 1 and args.gpus > 1:
         raise SystemExit("multi-GPU bench must be launched via torchrun")
     comm = Comm(rank, world)
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", torch.cuda.current_device()) if use_cuda else torch.device("cpu")
+    if rank == 0 and world > 1:
+        print(f"[preflight] world={world} backend="
+              f"{dist.get_backend() if dist.is_initialized() else 'none'} "
+              f"rccl_env={rccl_env}", file=sys.stderr, flush=True)
 
     cfg = compose(config_yaml_dir(), "base", [f"llm_config={args.model}"])
     llm = cfg["llm_config"].to_plain()
@@ -132,9 +161,20 @@ def main() -> None:
     sync()
     elapsed = time.time() - t0
     # MAX over ranks (all ranks hit the same barriers; elapsed is rank-local)
-    elapsed = max(comm.all_gather_scalars(elapsed))
+    per_rank = comm.all_gather_scalars(elapsed)
+    elapsed = max(per_rank)
 
     if rank == 0:
+        if world > 1:
+            # per-rank spread: DVFS imbalance / straggler diagnosis for the
+            # scaling run (NOTES_NEXT_ROUND §Scaling)
+            ms = [e / args.steps * 1000.0 for e in per_rank]
+            print(json.dumps({
+                "per_rank_ms_per_step": [round(x, 2) for x in ms],
+                "rank_spread_pct": round(
+                    (max(ms) - min(ms)) / max(ms) * 100.0, 2
+                ),
+            }), file=sys.stderr, flush=True)
         total_tokens = tokens_per_step * args.steps * world
         result = {
             "metric": "tokens_per_sec_whole_node",

@@ -113,6 +113,15 @@ class FedServer:
         self.aggregate_momenta = bool(fl.get("aggregate_momenta", False))
         self.client_m1 = self.layout.like() if self.aggregate_momenta else None
         self.client_m2 = self.layout.like() if self.aggregate_momenta else None
+        # hung-rank watchdog (photon.fit_timeout_s; SURVEY §7 hard-part 3):
+        # opt-in TCPStore control plane beside the RCCL data plane
+        self.fit_timeout_s = float(cfg["photon"].get("fit_timeout_s", 0) or 0)
+        self.watchdog = None
+        self._rebuild_gen = 0
+        if self.fit_timeout_s > 0 and comm.is_distributed:
+            from .watchdog import RoundWatchdog
+
+            self.watchdog = RoundWatchdog(comm.rank, comm.world_size)
 
     # -- initialization / resume -------------------------------------------
     def initialize(self) -> None:
@@ -210,6 +219,16 @@ class FedServer:
         per_client_sq_norms: list[tuple[float, float]] = []  # (n_i, ||g_i||^2)
 
         t_fit = time.time()
+        suicide = None
+        if self.watchdog is not None:
+            # a rank whose own fit hangs must die so peers can proceed
+            # (reference worker auto_terminate, worker.py:437-448)
+            from .watchdog import fit_suicide_timer
+
+            suicide = fit_suicide_timer(
+                self.fit_timeout_s * max(len(my_cids), 1) + 30.0,
+                self.comm.rank,
+            )
         for cid in my_cids:
             try:
                 local_payload, n_samples, metrics = self.client.fit(
@@ -234,9 +253,43 @@ class FedServer:
                 if self.comm.rank == 0 or True:
                     print(f"[fed] client {cid} fit failed: {e!r}")
         fit_time = time.time() - t_fit
+        if suicide is not None:
+            suicide.cancel()
 
-        # agree on failures across ranks
+        # hung-rank detection BEFORE any collective: a dead rank must not
+        # deadlock the all-reduce. Survivors agree on the alive set via the
+        # TCPStore, rebuild the process group without the dead ranks, and
+        # count their clients as failures.
+        dead_clients = 0
+        if self.watchdog is not None:
+            import torch.distributed as dist
+
+            from .watchdog import rebuild_process_group
+
+            self.watchdog.report_fit_done(server_round)
+            alive = self.watchdog.agree_alive(server_round, self.fit_timeout_s)
+            if len(alive) < self.comm.world_size:
+                dead = sorted(set(range(self.comm.world_size)) - set(alive))
+                dead_clients = sum(len(assignment[r]) for r in dead)
+                backend = dist.get_backend()
+                self._rebuild_gen += 1
+                new_rank, new_world = rebuild_process_group(
+                    alive, self.comm.rank, backend,
+                    generation=self._rebuild_gen,
+                )
+                print(f"[watchdog] round {server_round}: ranks {dead} dead; "
+                      f"rebuilt group as rank {new_rank}/{new_world}",
+                      flush=True)
+                self.comm.rank, self.comm.world_size = new_rank, new_world
+                # the watchdog itself keeps OLD rank numbering for its store
+                # keys; re-key it to the new group
+                self.watchdog.rank = new_rank
+                self.watchdog.world_size = new_world
+
+        # agree on failures across ranks (+ dead ranks' clients, identical
+        # on every survivor so added after the sum)
         fail_total = sum(self.comm.all_gather_scalars(float(failures)))
+        fail_total += dead_clients
         if fail_total > self.accept_failures_cnt and not self.ignore_failed_rounds:
             raise TooManyFailuresError(
                 f"round {server_round}: {int(fail_total)} client failures "
