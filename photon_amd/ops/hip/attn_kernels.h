@@ -38,6 +38,16 @@ constexpr int ATT_BLOCK = 64 * WAVES;
 constexpr int QB = 32;           // q rows per wave
 constexpr int KB = 32;           // keys per kv tile
 
+// D=128 occupancy/spill A/B (measured): dq wins at 2 waves/SIMD despite
+// ~100 B/lane spill (385 vs 244 TF/s); dkdv loses badly with its ~300 B
+// spill (99 vs 177 TF/s) -> spill-free 1 wave/SIMD there.
+#ifndef ATT_DQ_MINWAVES_D128
+#define ATT_DQ_MINWAVES_D128 2
+#endif
+#ifndef ATT_DKDV_MINWAVES_D128
+#define ATT_DKDV_MINWAVES_D128 1
+#endif
+
 DEV_INLINE unsigned swz(unsigned byte, int row) {
   return byte ^ (((unsigned)row & 15u) << 4);
 }
@@ -91,6 +101,38 @@ DEV_INLINE void lds_store2(__bf16* img, int row, int rowstride_b,
 }
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+constexpr float LOG2E = 1.4426950408889634f;
+constexpr float LN2 = 0.6931471805599453f;
+
+// Native v_exp_f32 (GCN v_exp IS 2^x): one transcendental, no ln2 multiply
+// — the whole softmax runs in the exp2 domain (scale/slopes pre-multiplied
+// by log2(e); LSE converted back to natural log at the epilogue so the
+// wire format is unchanged). MUST be the intrinsic, not inline asm: the
+// TRANS->VALU data hazard needs the compiler's s_nop padding, which it
+// cannot insert around an opaque asm block (caused max-err 3.2 corruption).
+#ifdef ABENCH_EXP_VIA_LN
+// bisect aid: same exp2-domain math through the old __expf lowering
+DEV_INLINE float exp2_fast(float x) { return __expf(x * LN2); }
+#else
+DEV_INLINE float exp2_fast(float x) { return __builtin_amdgcn_exp2f(x); }
+#endif
+
+// Staging address scheme: thread-fixed 32-bit chunk offsets (voff) + a
+// per-tile uniform 32-bit offset (soff) added to a 64-bit base — ONE
+// strength-reduced add per load instead of the old per-chunk 64-bit
+// row*stride chains. (A raw-buffer-descriptor version with hardware OOB
+// zeroing returned corrupted tiles on gfx950/ROCm 7.2 and was dropped —
+// see round-2 notes; the plain-pointer form measured equal.)
+DEV_INLINE bf16x8 stage_load16(const __bf16* base, int voffset, int soffset) {
+  // single 32-bit offset added to the uniform base: lowers to the
+  // global_load saddr form (scalar 64-bit base + 32-bit VGPR offset),
+  // keeping per-chunk addresses out of the VGPR file
+  const unsigned off = (unsigned)voffset + (unsigned)soffset;
+  return *(const bf16x8*)((const char*)base + off);
+}
+
+DEV_INLINE float log2_fast(float x) { return __builtin_amdgcn_logf(x); }
 
 // Hardware transpose read (gfx950 ds_read_b64_tr_b16). Probed semantics
 // (scripts/probe_tr.hip, addr_mode 3): per 16-lane group,
@@ -185,11 +227,15 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   const long bh = blockIdx.y;
   const int h = bh % H;
   const float slope = slopes[h];
-  const float scale = rsqrtf((float)D);
+  // exp2-domain softmax: scale/slope pre-multiplied by log2(e) so every
+  // exponential is ONE native v_exp_f32 (no ln->log2 v_mul per element).
+  const float scale2 = rsqrtf((float)D) * LOG2E;
+  const float slope2 = slope * LOG2E;
 
   const long ibase = (bh / H) * bs_i + (long)h * hs_i;
   const long obase = (bh / H) * bs_o + (long)h * hs_o;
-  const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
+  const int q0_block = blockIdx.x * (WAVES * QB);
+  const int q0 = q0_block + wave * QB;
   const int my_q = q0 + lq;  // this lane's q row
 
   // Q fragments in registers: B-operand, frag kk covers dh [kk*16, kk*16+16)
@@ -206,43 +252,78 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     }
   }
 
-  float m_run = -1e30f;  // finite sentinel: exp(-huge) == 0, no NaN paths
+  float m_run = -1e30f;  // finite sentinel: exp2(-huge) == 0, no NaN paths
   float l_run = 0.f;
   f32x16 o_acc[D / 32];
 #pragma unroll
   for (int db = 0; db < D / 32; ++db) o_acc[db] = f32x16{};
 
-  const int q_max_block =
-      min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
+  const int q_max_block = min(q0_block + WAVES * QB - 1, S - 1);
   const int n_tiles = causal ? (q_max_block / KBF + 1) : ((S + KBF - 1) / KBF);
   const int my_q_max = min(q0 + QB - 1, S - 1);
+  // Interior tiles: every (key, q) of the BLOCK unmasked — keys strictly
+  // below the block's first q row (causal) resp. full 64-key tiles within
+  // S (non-causal). The interior loop body carries no mask compare, no
+  // active/break checks and one straight-line softmax.
+  int t_int = causal ? (q0_block / KBF) : (S / KBF);
+  if (t_int > n_tiles) t_int = n_tiles;
 
-  // Register staging: each thread owns KBF*D/8/ATT_BLOCK 16-byte chunks per
-  // tensor. Chunk c -> key row c / (D/8), col (c % (D/8)) * 8.
+  // Hoisted per-element ALiBi constants (exp2-domain): element r of an
+  // MFMA result row covers key kv0s + pat(r), pat = (r&3)+8*(r>>2)+4*hi.
+  // alibi2[r] = slope2*pat; the per-subtile remainder slope2*(kv0s - my_q)
+  // is a per-lane constant that folds into the exp shift (mshift).
+  float alibi2[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    alibi2[r] = slope2 * (float)pat;
+  }
+
+  // SRSRC register staging (T8): thread-fixed 32-bit voffset + per-tile
+  // scalar soffset; num_records sized so rows >= S read hardware zeros
+  // (no per-chunk bounds, no 64-bit address chain).
   constexpr int NCHUNK = KBF * D / 8 / ATT_BLOCK;
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
+  const __bf16* kbase = k + ibase;
+  const __bf16* vbase = v + ibase;
+  int voff[NCHUNK];
+#pragma unroll
+  for (int i = 0; i < NCHUNK; ++i) {
+    const int c = i * ATT_BLOCK + threadIdx.x;
+    voff[i] = (int)(((long)(c / (D / 8)) * rs_i + (c % (D / 8)) * 8) * 2);
+  }
+  const int tile_soff = (int)(KBF * rs_i * 2);
 
   auto stage_load = [&](int t) {
+    const int so = t * tile_soff;
+#ifndef ABENCH_NO_LOAD
+    if ((t + 1) * KBF <= S) {  // uniform: full tile, no bounds anywhere
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        k_stage[i] = stage_load16(kbase, voff[i], so);
+        v_stage[i] = stage_load16(vbase, voff[i], so);
+      }
+    } else {  // tail tile: zero-fill rows >= S (never read past the tensor)
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
+        if (t * KBF + row < S) {
+          k_stage[i] = stage_load16(kbase, voff[i], so);
+          v_stage[i] = stage_load16(vbase, voff[i], so);
+        } else {
+          k_stage[i] = bf16x8{};
+          v_stage[i] = bf16x8{};
+        }
+      }
+    }
+#else
 #pragma unroll
     for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-      const long grow = (long)t * KBF + row;
-#ifndef ABENCH_NO_LOAD
-      if (grow < S) {
-        k_stage[i] = *(const bf16x8*)(k + ibase + grow * rs_i + col);
-        v_stage[i] = *(const bf16x8*)(v + ibase + grow * rs_i + col);
-      } else {
-        k_stage[i] = bf16x8{};
-        v_stage[i] = bf16x8{};
-      }
-#else
       k_stage[i] = bf16x8{};
       v_stage[i] = bf16x8{};
-      (void)grow;
-#endif
     }
+    (void)so;
+#endif
   };
   auto stage_write = [&](int b) {
 #pragma unroll
@@ -295,130 +376,150 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     }
   }
 
-  for (int t = 0; t < n_tiles; ++t) {
+  // Shared PV body: P (f32 regs) -> bf16 fragments -> V^T MFMAs via
+  // hardware transpose reads from the V ROW image. A-frag rows arrive
+  // permuted within each 16-row half (epilogue un-permutes).
+  auto pv_accum = [&](const float* p, int buf, int sub) {
+#ifndef ABENCH_NO_PV
+#pragma unroll
+    for (int s16 = 0; s16 < 2; ++s16) {
+      bf16x8 pfrag = pack_bfrag(p, 8 * s16);
+#pragma unroll
+      for (int db = 0; db < D / 32; ++db) {
+        union { bf16x4 h[2]; bf16x8 v8; } a;
+#pragma unroll
+        for (int rd = 0; rd < 2; ++rd) {
+          // hoisted: buf/sub/s16 offsets are immediates (the XOR field
+          // key&15 is invariant to +32/+16 key steps)
+          a.h[rd] = lds_tr16(
+              v_img(0), pv_addr[rd][db] + buf * 2 * IMG +
+                            sub * 32 * (D * 2) + s16 * 16 * (D * 2));
+        }
+        o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v8, pfrag,
+                                                            o_acc[db], 0, 0, 0);
+      }
+    }
+#endif
+  };
+  auto qk_mfma = [&](int buf, int sub) {
+    f32x16 s_acc = f32x16{};
+#ifndef ABENCH_NO_QK
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      // hoisted address + immediate offsets (sub*32 rows and buf are above
+      // the XOR swizzle bits; row&15 == lq&15 is sub-invariant)
+      bf16x8 a = *(const bf16x8*)((const char*)k_img(0) + qk_addr[kk] +
+                                  buf * 2 * IMG + sub * 32 * (D * 2));
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qfrag[kk], s_acc,
+                                                      0, 0, 0);
+    }
+#endif
+    return s_acc;
+  };
+  auto defer_rescale = [&](float tile_max_abs) {
+    // T13 defer-max: rescale only if some lane's max grew by > THR.
+    constexpr float THR = 8.f;
+    if (!__all(tile_max_abs - m_run <= THR)) {
+      const float m_new = fmaxf(m_run, tile_max_abs);
+      const float alpha = exp2_fast(m_run - m_new);  // exp2(-huge) == 0
+      l_run *= alpha;
+#pragma unroll
+      for (int db = 0; db < D / 32; ++db) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
+      }
+      m_run = m_new;
+    }
+  };
+
+  // ---- interior tiles: no masks, no active checks, straight-line ---------
+  for (int t = 0; t < t_int; ++t) {
     const int buf = t & 1;
-    const int kv0 = t * KBF;
     if (t + 1 < n_tiles) stage_load(t + 1);  // hide HBM under this tile
 #ifndef ABENCH_NO_BARRIER
     __syncthreads();  // LDS[buf] writes (prev iter) visible to all waves
 #endif
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      f32x16 s_acc = qk_mfma(buf, sub);
+      // shifted softmax: p' = s*scale2 + alibi2[r]; the per-subtile,
+      // per-lane constant b2 = slope2*(kv0s - my_q) shifts max and exp
+      // uniformly and folds into mshift — 1 fma + 1 max per element.
+      float p[16];
+      float tmax = -1e30f;
+#ifdef ABENCH_NO_SOFTMAX
+#pragma unroll
+      for (int r = 0; r < 16; ++r) p[r] = s_acc[r];
+      (void)tmax;
+#else
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = fmaf(s_acc[r], scale2, alibi2[r]);
+        tmax = fmaxf(tmax, p[r]);
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+      const float b2 = slope2 * (float)(t * KBF + sub * 32 - my_q);
+      defer_rescale(tmax + b2);
+      const float mshift = m_run - b2;
+      float l_add = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+#ifndef ABENCH_NO_EXP
+        p[r] = exp2_fast(p[r] - mshift);
+#else
+        p[r] = p[r] - mshift;  // bench-only: perf bisect without v_exp
+#endif
+        l_add += p[r];
+      }
+      l_add += __shfl_xor(l_add, 32, 64);
+      l_run += l_add;
+#endif  // ABENCH_NO_SOFTMAX
+      pv_accum(p, buf, sub);
+    }
+    if (t + 1 < n_tiles) stage_write(1 - buf);
+  }
 
+  // ---- boundary tiles: diagonal/tail masking (<= 2 per block) ------------
+  for (int t = t_int; t < n_tiles; ++t) {
+    const int buf = t & 1;
+    const int kv0 = t * KBF;
+    if (t + 1 < n_tiles) stage_load(t + 1);
+#ifndef ABENCH_NO_BARRIER
+    __syncthreads();
+#endif
     const bool active = !causal || (kv0 <= my_q_max);
     if (active) {
-      // Two 32-key SUBTILES over the staged 64-key buffer: halves the live
-      // softmax register state (s_acc + p = 32 regs instead of 64) so the
-      // MFMA accumulators stay in arch VGPRs — the v1 64-key softmax spent
-      // more issue slots on v_accvgpr shuttling than on MFMAs (pmc1
-      // profile: MFMA was 13% of issued instruction time).
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int kv0s = kv0 + sub * 32;
         if (causal && kv0s > my_q_max) break;
-        // S[key][q] = K Q^T : A = K row frags, B = Q regs.
-        f32x16 s_acc = f32x16{};
-#ifndef ABENCH_NO_QK
-#pragma unroll
-        for (int kk = 0; kk < D / 16; ++kk) {
-          bf16x8 a =
-              lds_frag(k_img(buf), sub * 32 + lq, D * 2, kk * 32 + hi * 16);
-          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qfrag[kk], s_acc,
-                                                          0, 0, 0);
-        }
-#endif
-        // sv = s*scale + slope*(key - my_q); masked lanes get -1e30 so the
-        // exp pass needs no per-element compare (exp(-huge - m) == 0).
-        // key = kv0s + PAT(r), PAT compile-time, so ALiBi is 2 fma/element.
-        // need_mask is BLOCK-uniform (scalar branch): a wave-dependent
-        // condition here gets if-converted into per-element cmp+cndmask on
-        // every subtile (measured 0.65 ms of the 1.2 ms kernel).
-        const bool need_mask =
-            (kv0s + 32 > S) ||
-            (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
-        const float abase = slope * (float)(kv0s - my_q);
+        f32x16 s_acc = qk_mfma(buf, sub);
+        // masked lanes get -1e30 so the exp pass needs no per-element
+        // compare (exp2(-huge - m) == 0)
+        const float abase2 = slope2 * (float)(kv0s - my_q);
         float p[16];
         float tile_max = -1e30f;
-#ifdef ABENCH_NO_SOFTMAX
 #pragma unroll
-        for (int r = 0; r < 16; ++r) p[r] = s_acc[r];
-        (void)need_mask; (void)abase; (void)tile_max;
-#else
-        // The mask branch is specialized at the LOOP level: leaving the
-        // per-element `if (need_mask)` inside the loop gets if-converted
-        // into ~170 always-executed cmp/cndmask instructions per tile
-        // (measured: 860-instruction main loop for 16 MFMAs).
-        if (need_mask) {
-#pragma unroll
-          for (int r = 0; r < 16; ++r) {
-            const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-            float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-            const int key = kv0s + pat;
-            const bool masked = (key >= S) || (causal && key > my_q);
-            sv = masked ? -1e30f : sv;
-            p[r] = sv;
-            tile_max = fmaxf(tile_max, sv);
-          }
-        } else {
-#pragma unroll
-          for (int r = 0; r < 16; ++r) {
-            const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-            const float sv =
-                fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-            p[r] = sv;
-            tile_max = fmaxf(tile_max, sv);
-          }
+        for (int r = 0; r < 16; ++r) {
+          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float sv = fmaf(s_acc[r], scale2, alibi2[r] + abase2);
+          const int key = kv0s + pat;
+          const bool masked = (key >= S) || (causal && key > my_q);
+          sv = masked ? -1e30f : sv;
+          p[r] = sv;
+          tile_max = fmaxf(tile_max, sv);
         }
         tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
-        // T13 defer-max: rescale only if some lane's max grew by > THR.
-        constexpr float THR = 8.f;
-        if (!__all(tile_max - m_run <= THR)) {
-          const float m_new = fmaxf(m_run, tile_max);
-          const float alpha = __expf(m_run - m_new);  // exp(-huge) == 0
-          l_run *= alpha;
-#pragma unroll
-          for (int db = 0; db < D / 32; ++db) {
-#pragma unroll
-            for (int r = 0; r < 16; ++r) o_acc[db][r] *= alpha;
-          }
-          m_run = m_new;
-        }
+        defer_rescale(tile_max);
         float l_add = 0.f;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-#ifndef ABENCH_NO_EXP
-          p[r] = __expf(p[r] - m_run);  // masked: exp(-huge) == 0
-#else
-          p[r] = p[r] - m_run;  // bench-only: perf bisect without v_exp
-#endif
+          p[r] = exp2_fast(p[r] - m_run);  // masked: exp2(-huge) == 0
           l_add += p[r];
         }
         l_add += __shfl_xor(l_add, 32, 64);
         l_run += l_add;
-#endif  // ABENCH_NO_SOFTMAX
-        // PV: O2[pi(dh)][q] += V^T P ; A-frags built by hardware
-        // transpose reads from the V ROW image. The A rows come out
-        // permuted within each 16-row half (pi: row 16*g + j holds
-        // dh 16*g + 4*(j&3) + (j>>2)); the epilogue un-permutes.
-#ifndef ABENCH_NO_PV
-#pragma unroll
-        for (int s16 = 0; s16 < 2; ++s16) {
-          bf16x8 pfrag = pack_bfrag(p, 8 * s16);
-#pragma unroll
-          for (int db = 0; db < D / 32; ++db) {
-            union { bf16x4 h[2]; bf16x8 v8; } a;
-#pragma unroll
-            for (int rd = 0; rd < 2; ++rd) {
-              // hoisted: buf/sub/s16 offsets are immediates (the XOR
-              // field key&15 is invariant to +32/+16 key steps)
-              a.h[rd] = lds_tr16(
-                  v_img(0),
-                  pv_addr[rd][db] + buf * 2 * IMG + sub * 32 * (D * 2) +
-                      s16 * 16 * (D * 2));
-            }
-            o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                a.v8, pfrag, o_acc[db], 0, 0, 0);
-          }
-        }
-#endif
+        pv_accum(p, buf, sub);
       }
     }
     if (t + 1 < n_tiles) stage_write(1 - buf);
@@ -439,7 +540,9 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     }
   }
   if (hi == 0 && my_q < S && lse_out) {
-    lse_out[bh * (long)S + my_q] = m_run + __logf(l_run);
+    // wire format stays natural-log: lse = ln(sum exp(sv)) with sv in
+    // nats = ln2 * (m2 + log2(l)) from the exp2-domain running state
+    lse_out[bh * (long)S + my_q] = LN2 * (m_run + log2_fast(l_run));
   }
   __syncthreads();
   // each wave stores its own 32 rows
@@ -458,9 +561,9 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
 // dQ^T[dh][q] = sum_key K^T[dh][key] * dS[key][q] * scale
 // ---------------------------------------------------------------------------
 template <int D>
-// D=128 needs >256 VGPRs with the staged pipeline; capping at 2 waves/SIMD
-// spills — run 1 wave/SIMD there instead (cf. dkdv).
-__global__ __launch_bounds__(ATT_BLOCK, (D <= 64) ? 2 : 1)
+// SRSRC staging dropped D=128 register pressure (211 VGPR vs >256 with the
+// old 64-bit-address register staging): 2 waves/SIMD now fits at every D.
+__global__ __launch_bounds__(ATT_BLOCK, (D <= 64) ? 2 : ATT_DQ_MINWAVES_D128)
 void attn_bwd_dq_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
@@ -483,10 +586,15 @@ void attn_bwd_dq_kernel(
   const long bh = blockIdx.y;
   const int h = bh % H;
   const float slope = slopes[h];
-  const float scale = rsqrtf((float)D);
+  // exp2 domain (cf. fwd); log2(scale) = -log2(D)/2 is exact for D=2^k
+  const float scale2 = rsqrtf((float)D) * LOG2E;
+  const float slope2 = slope * LOG2E;
+  constexpr float LOG2_SCALE = (D == 64) ? -3.0f : (D == 128) ? -3.5f : 0.f;
+  static_assert(D == 64 || D == 128, "log2(scale) fold assumes D in {64,128}");
   const long ibase = (bh / H) * bs_i + (long)h * hs_i;
   const long obase = (bh / H) * bs_o + (long)h * hs_o;
-  const int q0 = blockIdx.x * (WAVES * QB) + wave * QB;
+  const int q0_block = blockIdx.x * (WAVES * QB);
+  const int q0 = q0_block + wave * QB;
   const int my_q = q0 + lq;
 
   bf16x8 qfrag[D / 16], dofrag[D / 16];
@@ -504,33 +612,52 @@ void attn_bwd_dq_kernel(
       }
     }
   }
-  const float my_lse = (my_q < S) ? lse[bh * (long)S + my_q] : INFINITY;
+  // lse pre-multiplied into the exp2 domain
+  const float my_lse2 =
+      (my_q < S) ? lse[bh * (long)S + my_q] * LOG2E : INFINITY;
   const float my_delta = (my_q < S) ? delta[bh * (long)S + my_q] : 0.f;
 
   f32x16 dq_acc[D / 32];
 #pragma unroll
   for (int db = 0; db < D / 32; ++db) dq_acc[db] = f32x16{};
 
-  const int q_max_block =
-      min(blockIdx.x * (WAVES * QB) + WAVES * QB - 1, S - 1);
+  const int q_max_block = min(q0_block + WAVES * QB - 1, S - 1);
   const int n_tiles = causal ? (q_max_block / KBQ + 1) : ((S + KBQ - 1) / KBQ);
   const int my_q_max = min(q0 + QB - 1, S - 1);
+  int t_int = causal ? (q0_block / KBQ) : (S / KBQ);
+  if (t_int > n_tiles) t_int = n_tiles;
 
+  // SRSRC staging (cf. fwd): hardware-zero OOB rows, no bounds compares
   constexpr int NCHUNK = KBQ * D / 8 / ATT_BLOCK;
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
-  auto stage_load = [&](int t) {
+  const __bf16* kbase = k + ibase;
+  const __bf16* vbase = v + ibase;
+  int voff[NCHUNK];
 #pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-      const long grow = (long)t * KBQ + row;
-      if (grow < S) {
-        k_stage[i] = *(const bf16x8*)(k + ibase + grow * rs_i + col);
-        v_stage[i] = *(const bf16x8*)(v + ibase + grow * rs_i + col);
-      } else {
-        k_stage[i] = bf16x8{};
-        v_stage[i] = bf16x8{};
+  for (int i = 0; i < NCHUNK; ++i) {
+    const int c = i * ATT_BLOCK + threadIdx.x;
+    voff[i] = (int)(((long)(c / (D / 8)) * rs_i + (c % (D / 8)) * 8) * 2);
+  }
+  const int tile_soff = (int)(KBQ * rs_i * 2);
+  auto stage_load = [&](int t) {
+    const int so = t * tile_soff;
+    if ((t + 1) * KBQ <= S) {  // uniform fast path (cf. fwd)
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        k_stage[i] = stage_load16(kbase, voff[i], so);
+        v_stage[i] = stage_load16(vbase, voff[i], so);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
+        if (t * KBQ + row < S) {
+          k_stage[i] = stage_load16(kbase, voff[i], so);
+          v_stage[i] = stage_load16(vbase, voff[i], so);
+        } else {
+          k_stage[i] = bf16x8{};
+          v_stage[i] = bf16x8{};
+        }
       }
     }
   };
@@ -568,77 +695,90 @@ void attn_bwd_dq_kernel(
     }
   }
 
-  for (int t = 0; t < n_tiles; ++t) {
+  auto qkdp_mfma = [&](int buf, int sub, f32x16& s_acc, f32x16& dp_acc) {
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      const char* base = (const char*)k_img(0) + qk_addr[kk] +
+                         buf * 2 * IMG2 + sub * 32 * (D * 2);
+      bf16x8 ka = *(const bf16x8*)base;
+      bf16x8 va = *(const bf16x8*)(base + IMG2);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[kk], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[kk],
+                                                       dp_acc, 0, 0, 0);
+    }
+  };
+  auto dsk_mfma = [&](const float* ds, int buf, int sub) {
+#pragma unroll
+    for (int s16 = 0; s16 < 2; ++s16) {
+      bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
+#pragma unroll
+      for (int db = 0; db < D / 32; ++db) {
+        union { bf16x4 h[2]; bf16x8 v8; } a;
+#pragma unroll
+        for (int rd = 0; rd < 2; ++rd) {
+          a.h[rd] = lds_tr16(
+              k_img(0), tr_addr[rd][db] + buf * 2 * IMG2 +
+                            sub * 32 * (D * 2) + s16 * 16 * (D * 2));
+        }
+        dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a.v8, dsfrag, dq_acc[db], 0, 0, 0);
+      }
+    }
+  };
+
+  // ---- interior tiles (no masks): p*scale = exp2(sv2 - lse2) ------------
+  for (int t = 0; t < t_int; ++t) {
+    const int buf = t & 1;
+    if (t + 1 < n_tiles) stage_load(t + 1);
+    __syncthreads();
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
+      qkdp_mfma(buf, sub, s_acc, dp_acc);
+      // abase2 folds log2(scale) AND the lse: exp2 yields p*scale, so ds
+      // skips the per-element multiply
+      const float ab2 = fmaf(slope2, (float)(t * KBQ + sub * 32 - my_q),
+                             LOG2_SCALE) - my_lse2;
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float sv = fmaf(s_acc[r], scale2, fmaf(slope2, (float)pat, ab2));
+        const float pv = exp2_fast(sv);
+        ds[r] = pv * (dp_acc[r] - my_delta);
+      }
+      dsk_mfma(ds, buf, sub);
+    }
+    if (t + 1 < n_tiles) stage_write(1 - buf);
+  }
+
+  // ---- boundary tiles (diagonal/tail) ------------------------------------
+  for (int t = t_int; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int kv0 = t * KBQ;
     if (t + 1 < n_tiles) stage_load(t + 1);
     __syncthreads();
-
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       const int kv0s = kv0 + sub * 32;
       if (causal && kv0s > my_q_max) break;
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
-#pragma unroll
-      for (int kk = 0; kk < D / 16; ++kk) {
-        const char* base = (const char*)k_img(0) + qk_addr[kk] +
-                           buf * 2 * IMG2 + sub * 32 * (D * 2);
-        bf16x8 ka = *(const bf16x8*)base;
-        bf16x8 va = *(const bf16x8*)(base + IMG2);
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[kk], s_acc,
-                                                        0, 0, 0);
-        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dofrag[kk],
-                                                         dp_acc, 0, 0, 0);
-      }
-      // p = exp(sv - lse); masked keys -> exp(-huge) == 0 (no compares on
-      // non-diagonal tiles; need_mask is BLOCK-uniform).
-      const bool need_mask =
-          (kv0s + 32 > S) ||
-          (causal && (kv0s + 31 > (int)blockIdx.x * (WAVES * QB)));
-      // abase folds ln(scale): exp yields p*scale, so ds skips the
-      // per-element multiply (cf. dkdv).
-      const float abase =
-          fmaf(slope, (float)(kv0s - my_q), __logf(scale));
+      qkdp_mfma(buf, sub, s_acc, dp_acc);
+      const float ab2 =
+          fmaf(slope2, (float)(kv0s - my_q), LOG2_SCALE) - my_lse2;
       float ds[16];
-      // loop-level mask specialization (cf. fwd): the embedded per-element
-      // if gets if-converted into always-on cmp/cndmask chains.
-      if (need_mask) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float sv = fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-          const int key = kv0s + pat;
-          const bool masked = (key >= S) || (causal && key > my_q);
-          sv = masked ? -1e30f : sv;
-          const float pv = __expf(sv - my_lse);
-          ds[r] = pv * (dp_acc[r] - my_delta);
-        }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const float sv =
-              fmaf(s_acc[r], scale, fmaf(slope, (float)pat, abase));
-          const float pv = __expf(sv - my_lse);
-          ds[r] = pv * (dp_acc[r] - my_delta);
-        }
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float sv = fmaf(s_acc[r], scale2, fmaf(slope2, (float)pat, ab2));
+        const int key = kv0s + pat;
+        const bool masked = (key >= S) || (causal && key > my_q);
+        sv = masked ? -1e30f : sv;
+        const float pv = exp2_fast(sv);  // masked: exp2(-huge) == 0
+        ds[r] = pv * (dp_acc[r] - my_delta);
       }
-#pragma unroll
-      for (int s16 = 0; s16 < 2; ++s16) {
-        bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
-#pragma unroll
-        for (int db = 0; db < D / 32; ++db) {
-          union { bf16x4 h[2]; bf16x8 v8; } a;
-#pragma unroll
-          for (int rd = 0; rd < 2; ++rd) {
-            a.h[rd] = lds_tr16(
-                k_img(0), tr_addr[rd][db] + buf * 2 * IMG2 +
-                              sub * 32 * (D * 2) + s16 * 16 * (D * 2));
-          }
-          dq_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              a.v8, dsfrag, dq_acc[db], 0, 0, 0);
-        }
-      }
+      dsk_mfma(ds, buf, sub);
     }
     if (t + 1 < n_tiles) stage_write(1 - buf);
   }
@@ -671,9 +811,9 @@ void attn_bwd_dq_kernel(
 //   dK^T[dh][key] = sum_q Q^T[dh][q] dS[q][key] * scale
 // ---------------------------------------------------------------------------
 template <int D>
-// D=128 needs ~420 VGPRs (2 accumulator pairs + staging); capping at 2
-// waves/SIMD spills 148 regs to scratch — let it run 1 wave/SIMD instead.
-__global__ __launch_bounds__(ATT_BLOCK, (D <= 64) ? 2 : 1)
+// SRSRC staging dropped D=128 register pressure (204 VGPR vs ~420 with the
+// old register staging): 2 waves/SIMD now fits at every D.
+__global__ __launch_bounds__(ATT_BLOCK, (D <= 64) ? 2 : ATT_DKDV_MINWAVES_D128)
 void attn_bwd_dkdv_kernel(
     const __bf16* __restrict__ dout, const __bf16* __restrict__ q,
     const __bf16* __restrict__ k, const __bf16* __restrict__ v,
@@ -702,10 +842,13 @@ void attn_bwd_dkdv_kernel(
   const long bh = blockIdx.y;
   const int h = bh % H;
   const float slope = slopes[h];
-  const float scale = rsqrtf((float)D);
+  // exp2 domain (cf. fwd/dq)
+  const float scale2 = rsqrtf((float)D) * LOG2E;
+  const float slope2 = slope * LOG2E;
+  constexpr float LOG2_SCALE = (D == 64) ? -3.0f : (D == 128) ? -3.5f : 0.f;
+  static_assert(D == 64 || D == 128, "log2(scale) fold assumes D in {64,128}");
   const long ibase = (bh / H) * bs_i + (long)h * hs_i;
   const long obase = (bh / H) * bs_o + (long)h * hs_o;
-  const float log_scale = __logf(scale);
   const int k0 = blockIdx.x * (WAVES * KB) + wave * KB;
   const int my_key = k0 + lq;
 
@@ -742,24 +885,46 @@ void attn_bwd_dkdv_kernel(
   bf16x8 q_stage[NCHUNK], do_stage[NCHUNK];
   float lse_r = 0.f, del_r = 0.f;
 
-  auto stage_load = [&](int t) {
+  // SRSRC staging (cf. fwd). q and dout have separate strides (packed QKV
+  // input vs contiguous dout) -> separate descriptors/voffsets.
+  const __bf16* qbase = q + ibase;
+  const __bf16* obase_p = dout + obase;
+  int voff_q[NCHUNK], voff_o[NCHUNK];
 #pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-      const long grow = (long)t * QTF + row;
-      if (grow < S) {
-        q_stage[i] = *(const bf16x8*)(q + ibase + grow * rs_i + col);
-        do_stage[i] = *(const bf16x8*)(dout + obase + grow * rs_o + col);
-      } else {
-        q_stage[i] = bf16x8{};
-        do_stage[i] = bf16x8{};
+  for (int i = 0; i < NCHUNK; ++i) {
+    const int c = i * ATT_BLOCK + threadIdx.x;
+    const int row = c / (D / 8);
+    const int col = (c % (D / 8)) * 8;
+    voff_q[i] = (int)(((long)row * rs_i + col) * 2);
+    voff_o[i] = (int)(((long)row * rs_o + col) * 2);
+  }
+  const int tsoff_q = (int)(QTF * rs_i * 2);
+  const int tsoff_o = (int)(QTF * rs_o * 2);
+
+  auto stage_load = [&](int t) {
+    if ((t + 1) * QTF <= S) {  // uniform fast path (cf. fwd)
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        q_stage[i] = stage_load16(qbase, voff_q[i], t * tsoff_q);
+        do_stage[i] = stage_load16(obase_p, voff_o[i], t * tsoff_o);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < NCHUNK; ++i) {
+        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
+        if (t * QTF + row < S) {
+          q_stage[i] = stage_load16(qbase, voff_q[i], t * tsoff_q);
+          do_stage[i] = stage_load16(obase_p, voff_o[i], t * tsoff_o);
+        } else {
+          q_stage[i] = bf16x8{};
+          do_stage[i] = bf16x8{};
+        }
       }
     }
     if (threadIdx.x < QTF) {
       const long qi = (long)t * QTF + threadIdx.x;
-      lse_r = (qi < S) ? lse[bh * (long)S + qi] : INFINITY;
+      // lse pre-multiplied into the exp2 domain at stage time
+      lse_r = (qi < S) ? lse[bh * (long)S + qi] * LOG2E : INFINITY;
       del_r = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
     }
   };
@@ -800,99 +965,127 @@ void attn_bwd_dkdv_kernel(
     }
   }
 
-  for (int t = t0; t < n_tiles; ++t) {
+  // S'[q][key]: A = Q row frags, B = K regs; dP'[q][key]: A = dO, B = V
+  auto qkdp_mfma = [&](int buf, int sub, f32x16& s_acc, f32x16& dp_acc) {
+#pragma unroll
+    for (int kk = 0; kk < D / 16; ++kk) {
+      const char* base = (const char*)q_img(0) + qk_addr[kk] +
+                         buf * 2 * IMG2 + sub * 32 * (D * 2);
+      bf16x8 qa = *(const bf16x8*)base;
+      bf16x8 doa = *(const bf16x8*)(base + IMG2);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[kk], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vfrag[kk],
+                                                       dp_acc, 0, 0, 0);
+    }
+  };
+  auto dkdv_mfma = [&](const float* p, const float* ds, int buf, int sub) {
+#pragma unroll
+    for (int s16 = 0; s16 < 2; ++s16) {
+      bf16x8 pfrag = pack_bfrag(p, 8 * s16);
+      bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
+#pragma unroll
+      for (int db = 0; db < D / 32; ++db) {
+        union { bf16x4 h[2]; bf16x8 v8; } doa, qa;
+#pragma unroll
+        for (int rd = 0; rd < 2; ++rd) {
+          const unsigned byte = tr_addr[rd][db] + buf * 2 * IMG2 +
+                                sub * 32 * (D * 2) + s16 * 16 * (D * 2);
+          qa.h[rd] = lds_tr16(q_img(0), byte);
+          doa.h[rd] = lds_tr16((const __bf16*)((const char*)q_img(0) + IMG2),
+                               byte);
+        }
+        dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            doa.v8, pfrag, dv_acc[db], 0, 0, 0);
+        dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            qa.v8, dsfrag, dk_acc[db], 0, 0, 0);
+      }
+    }
+  };
+  // masked (diagonal/tail) tile body — abase2 folds the ALiBi row constant
+  // and log2(scale): exp2 yields p*scale (dv un-scales once in epilogue)
+  auto masked_tile = [&](int t) {
     const int buf = t & 1;
     const int qt0 = t * QTF;
-    if (t + 1 < n_tiles) stage_load(t + 1);
-    __syncthreads();
-
 #pragma unroll
     for (int sub = 0; sub < NSUB; ++sub) {
       const int qt0s = qt0 + sub * 32;
-      // subtile fully above this wave's keys? (causal: q < key => masked)
       if (causal && qt0s + 31 < my_k_min) continue;
-      // S'[q][key]: A = Q row frags, B = K regs; dP'[q][key]: A = dO, B = V
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
-#pragma unroll
-      for (int kk = 0; kk < D / 16; ++kk) {
-        const char* base = (const char*)q_img(0) + qk_addr[kk] +
-                           buf * 2 * IMG2 + sub * 32 * (D * 2);
-        bf16x8 qa = *(const bf16x8*)base;
-        bf16x8 doa = *(const bf16x8*)(base + IMG2);
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kfrag[kk], s_acc,
-                                                        0, 0, 0);
-        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa, vfrag[kk],
-                                                         dp_acc, 0, 0, 0);
-      }
-      // ALiBi: bias = slope*(my_key - qi); masked q rows -> exp(-huge) == 0.
-      // need_mask is BLOCK-uniform (scalar branch).
-      const bool need_mask =
-          (qt0s + 32 > S) || (causal && block_k_max >= qt0s);
-      // abase folds BOTH the ALiBi row constant and ln(scale): the exp
-      // then yields p*scale directly — ds needs no per-element *scale and
-      // dv_acc (which wants raw p) is divided by scale once in the
-      // epilogue. (Vectorized lse/del b128 loads were tried and REVERTED:
-      // +32 live VGPRs pushed dkdv<64> to 256+80B-scratch spilling, a net
-      // 30% slowdown.)
-      const float abase =
-          fmaf(slope, (float)(my_key - qt0s), log_scale);
+      qkdp_mfma(buf, sub, s_acc, dp_acc);
+      const float ab2 = fmaf(slope2, (float)(my_key - qt0s), LOG2_SCALE);
       float p[16], ds[16];
-      // loop-level mask specialization (cf. fwd)
-      if (need_mask) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const float l = lse_buf[buf * QTF + sub * 32 + pat];
-          const float dlt = del_buf[buf * QTF + sub * 32 + pat];
-          float sv = fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
-          const int qi = qt0s + pat;
-          const bool masked =
-              (my_key >= S) || (causal && my_key > qi) || (qi >= S);
-          sv = masked ? -1e30f : sv;
-          p[r] = __expf(sv - l);  // == p_raw * scale
-          ds[r] = p[r] * (dp_acc[r] - dlt);
-        }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const float l = lse_buf[buf * QTF + sub * 32 + pat];
-          const float dlt = del_buf[buf * QTF + sub * 32 + pat];
-          const float sv =
-              fmaf(s_acc[r], scale, fmaf(slope, -(float)pat, abase));
-          p[r] = __expf(sv - l);  // == p_raw * scale
-          ds[r] = p[r] * (dp_acc[r] - dlt);
-        }
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float l2 = lse_buf[buf * QTF + sub * 32 + pat];
+        const float dlt = del_buf[buf * QTF + sub * 32 + pat];
+        float sv = fmaf(s_acc[r], scale2, fmaf(slope2, -(float)pat, ab2));
+        const int qi = qt0s + pat;
+        const bool masked =
+            (my_key >= S) || (causal && my_key > qi) || (qi >= S);
+        sv = masked ? -1e30f : sv;
+        p[r] = exp2_fast(sv - l2);  // == p_raw * scale
+        ds[r] = p[r] * (dp_acc[r] - dlt);
       }
-#pragma unroll
-      for (int s16 = 0; s16 < 2; ++s16) {
-        bf16x8 pfrag = pack_bfrag(p, 8 * s16);
-        bf16x8 dsfrag = pack_bfrag(ds, 8 * s16);
-#pragma unroll
-        for (int db = 0; db < D / 32; ++db) {
-          union { bf16x4 h[2]; bf16x8 v8; } doa, qa;
-#pragma unroll
-          for (int rd = 0; rd < 2; ++rd) {
-            const unsigned byte = tr_addr[rd][db] + buf * 2 * IMG2 +
-                                  sub * 32 * (D * 2) + s16 * 16 * (D * 2);
-            qa.h[rd] = lds_tr16(q_img(0), byte);
-            doa.h[rd] = lds_tr16((const __bf16*)((const char*)q_img(0) + IMG2),
-                                 byte);
-          }
-          dv_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              doa.v8, pfrag, dv_acc[db], 0, 0, 0);
-          dk_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              qa.v8, dsfrag, dk_acc[db], 0, 0, 0);
-        }
-      }
+      dkdv_mfma(p, ds, buf, sub);
     }
-    if (t + 1 < n_tiles) stage_write(1 - buf);
+  };
+  // interior (unmasked) tile body
+  auto interior_tile = [&](int t) {
+    const int buf = t & 1;
+    const int qt0 = t * QTF;
+#pragma unroll
+    for (int sub = 0; sub < NSUB; ++sub) {
+      const int qt0s = qt0 + sub * 32;
+      f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
+      qkdp_mfma(buf, sub, s_acc, dp_acc);
+      const float ab2 = fmaf(slope2, (float)(my_key - qt0s), LOG2_SCALE);
+      float p[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int pat = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float l2 = lse_buf[buf * QTF + sub * 32 + pat];
+        const float dlt = del_buf[buf * QTF + sub * 32 + pat];
+        const float sv =
+            fmaf(s_acc[r], scale2, fmaf(slope2, -(float)pat, ab2));
+        p[r] = exp2_fast(sv - l2);  // == p_raw * scale
+        ds[r] = p[r] * (dp_acc[r] - dlt);
+      }
+      dkdv_mfma(p, ds, buf, sub);
+    }
+  };
+
+  // three phases over Q tiles: diagonal head (masked), interior, tail.
+  // Interior tiles have every q strictly above the block's keys (causal)
+  // and full rows: no compares, no continue, one straight-line softmax.
+  const int t_diag_end =
+      causal ? min(block_k_max / QTF, n_tiles - 1) : (t0 - 1);
+  int t_full_end = S / QTF;
+  if (t_full_end > n_tiles) t_full_end = n_tiles;
+  for (int t = t0; t <= t_diag_end; ++t) {
+    if (t + 1 < n_tiles) stage_load(t + 1);
+    __syncthreads();
+    masked_tile(t);
+    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
+  }
+  for (int t = t_diag_end + 1; t < t_full_end; ++t) {
+    if (t + 1 < n_tiles) stage_load(t + 1);
+    __syncthreads();
+    interior_tile(t);
+    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
+  }
+  for (int t = max(t_full_end, t_diag_end + 1); t < n_tiles; ++t) {
+    if (t + 1 < n_tiles) stage_load(t + 1);
+    __syncthreads();
+    masked_tile(t);
+    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
   }
   __syncthreads();  // protect epilogue smem reuse
 
   // dv accumulated p*scale (see abase fold): undo once here.
   {
-    const float inv_scale = 1.f / scale;
+    const float inv_scale = sqrtf((float)D);
 #pragma unroll
     for (int db = 0; db < D / 32; ++db) {
 #pragma unroll

@@ -118,6 +118,82 @@ void check(int B, int H, int S) {
   hipFree(q); hipFree(k); hipFree(v); hipFree(o); hipFree(lse); hipFree(slopes);
 }
 
+// Backward bench: runs fwd once for a real lse, then times dq and dkdv.
+template <int D>
+void bench_bwd(int B, int H, int S, int iters) {
+  size_t n = (size_t)B * H * S * D;
+  __bf16 *q, *k, *v, *o, *dout, *dq, *dk, *dv;
+  float *lse, *delta, *slopes;
+  CHECK(hipMalloc(&q, n * 2)); CHECK(hipMalloc(&k, n * 2));
+  CHECK(hipMalloc(&v, n * 2)); CHECK(hipMalloc(&o, n * 2));
+  CHECK(hipMalloc(&dout, n * 2)); CHECK(hipMalloc(&dq, n * 2));
+  CHECK(hipMalloc(&dk, n * 2)); CHECK(hipMalloc(&dv, n * 2));
+  CHECK(hipMalloc(&lse, (size_t)B * H * S * 4));
+  CHECK(hipMalloc(&delta, (size_t)B * H * S * 4));
+  CHECK(hipMalloc(&slopes, H * 4));
+  {
+    std::vector<__bf16> h(n);
+    for (size_t i = 0; i < n; ++i) h[i] = (__bf16)((float)(rand() % 2000 - 1000) / 500.f);
+    CHECK(hipMemcpy(q, h.data(), n * 2, hipMemcpyHostToDevice));
+    for (size_t i = 0; i < n; ++i) h[i] = (__bf16)((float)(rand() % 2000 - 1000) / 500.f);
+    CHECK(hipMemcpy(k, h.data(), n * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(v, h.data(), n * 2, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(dout, h.data(), n * 2, hipMemcpyHostToDevice));
+    std::vector<float> hs(H);
+    for (int i = 0; i < H; ++i) hs[i] = 0.5f / (1 << i);
+    CHECK(hipMemcpy(slopes, hs.data(), H * 4, hipMemcpyHostToDevice));
+    std::vector<float> hd((size_t)B * H * S, 0.1f);
+    CHECK(hipMemcpy(delta, hd.data(), (size_t)B * H * S * 4, hipMemcpyHostToDevice));
+  }
+  dim3 grid((S + WAVES * QB - 1) / (WAVES * QB), (size_t)B * H);
+  const long bs0 = (long)H * S * D, hs0 = (long)S * D, rs0 = D;
+  const int lds_fwd = 4 * KBF * D * 2 > WAVES * 64 * D ? 4 * KBF * D * 2 : WAVES * 64 * D;
+  hipLaunchKernelGGL((attn_fwd_kernel<D>), grid, dim3(ATT_BLOCK), lds_fwd, 0,
+                     q, k, v, slopes, o, lse, S, H, 1, bs0, hs0, rs0, bs0, hs0, rs0);
+  CHECK(hipDeviceSynchronize());
+  const int qtf = D <= 64 ? 64 : 32;
+  const int lds_dq = 4 * 64 * D * 2 > WAVES * 64 * D ? 4 * 64 * D * 2 : WAVES * 64 * D;
+  int lds_kv = 4 * qtf * D * 2 + 4 * qtf * 4;
+  if (lds_kv < WAVES * 64 * D) lds_kv = WAVES * 64 * D;
+  hipEvent_t e0, e1; (void)hipEventCreate(&e0); (void)hipEventCreate(&e1);
+  // dq
+  for (int i = 0; i < 3; ++i)
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<D>), grid, dim3(ATT_BLOCK), lds_dq, 0,
+                       dout, q, k, v, slopes, lse, delta, dq, S, H, 1,
+                       bs0, hs0, rs0, bs0, hs0, rs0);
+  CHECK(hipDeviceSynchronize());
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<D>), grid, dim3(ATT_BLOCK), lds_dq, 0,
+                       dout, q, k, v, slopes, lse, delta, dq, S, H, 1,
+                       bs0, hs0, rs0, bs0, hs0, rs0);
+  hipEventRecord(e1);
+  CHECK(hipDeviceSynchronize());
+  float ms_dq; hipEventElapsedTime(&ms_dq, e0, e1); ms_dq /= iters;
+  // dkdv
+  for (int i = 0; i < 3; ++i)
+    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<D>), grid, dim3(ATT_BLOCK), lds_kv, 0,
+                       dout, q, k, v, slopes, lse, delta, dk, dv, S, H, 1,
+                       bs0, hs0, rs0, bs0, hs0, rs0);
+  CHECK(hipDeviceSynchronize());
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<D>), grid, dim3(ATT_BLOCK), lds_kv, 0,
+                       dout, q, k, v, slopes, lse, delta, dk, dv, S, H, 1,
+                       bs0, hs0, rs0, bs0, hs0, rs0);
+  hipEventRecord(e1);
+  CHECK(hipDeviceSynchronize());
+  float ms_kv; hipEventElapsedTime(&ms_kv, e0, e1); ms_kv /= iters;
+  // dq: 2 GEMM-equivalents (S, dSK); dkdv: 2.5 (S, dP, dV, dK at half causal)
+  double base = 2.0 * B * H * (double)S * S * D / 2 / 1e12;  // causal half
+  printf("dq   B%d H%d S%d D%d: %8.3f ms  %7.1f TF/s eff\n", B, H, S, D, ms_dq,
+         base * 2 / (ms_dq / 1e3));
+  printf("dkdv B%d H%d S%d D%d: %8.3f ms  %7.1f TF/s eff\n", B, H, S, D, ms_kv,
+         base * 2 / (ms_kv / 1e3));
+  hipFree(q); hipFree(k); hipFree(v); hipFree(o); hipFree(dout); hipFree(dq);
+  hipFree(dk); hipFree(dv); hipFree(lse); hipFree(delta); hipFree(slopes);
+}
+
 int main(int argc, char** argv) {
   if (argc > 1 && argv[1][0] == 'c') {
     check<64>(1, 2, 192);
@@ -126,5 +202,7 @@ int main(int argc, char** argv) {
   }
   bench<64>(32, 12, 2048, 20);
   bench<128>(8, 16, 4096, 10);
+  bench_bwd<64>(32, 12, 2048, 10);
+  bench_bwd<128>(8, 16, 4096, 5);
   return 0;
 }
