@@ -124,13 +124,32 @@ DEV_INLINE float exp2_fast(float x) { return __builtin_amdgcn_exp2f(x); }
 // row*stride chains. (A raw-buffer-descriptor version with hardware OOB
 // zeroing returned corrupted tiles on gfx950/ROCm 7.2 and was dropped —
 // see round-2 notes; the plain-pointer form measured equal.)
-DEV_INLINE bf16x8 stage_load16(const __bf16* base, int voffset, int soffset) {
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#ifdef ATT_STAGE_RSRC2
+// Buffer-descriptor staging, descriptor REBUILT per tile (soffset folded
+// into the base; the SGPR-soffset form returned corrupted tiles on this
+// stack). num_records clamps rows >= S to hardware zeros.
+DEV_INLINE bf16x8 stage_load16x(const __bf16* base, long extent, int voffset,
+                                int soffset) {
+  auto rsrc = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)((const char*)base + (unsigned)soffset), 0,
+      (int)(extent - soffset), 0);
+  union { f32x4 f; bf16x8 b; } u;
+  u.f = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voffset, 0, 0);
+  return u.b;
+}
+#else
+DEV_INLINE bf16x8 stage_load16x(const __bf16* base, long extent, int voffset,
+                                int soffset) {
   // single 32-bit offset added to the uniform base: lowers to the
   // global_load saddr form (scalar 64-bit base + 32-bit VGPR offset),
   // keeping per-chunk addresses out of the VGPR file
+  (void)extent;
   const unsigned off = (unsigned)voffset + (unsigned)soffset;
   return *(const bf16x8*)((const char*)base + off);
 }
+#endif
 
 DEV_INLINE float log2_fast(float x) { return __builtin_amdgcn_logf(x); }
 
@@ -286,6 +305,7 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
   const __bf16* kbase = k + ibase;
   const __bf16* vbase = v + ibase;
+  const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
   int voff[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -300,16 +320,16 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     if ((t + 1) * KBF <= S) {  // uniform: full tile, no bounds anywhere
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
-        k_stage[i] = stage_load16(kbase, voff[i], so);
-        v_stage[i] = stage_load16(vbase, voff[i], so);
+        k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
+        v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
       }
     } else {  // tail tile: zero-fill rows >= S (never read past the tensor)
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
         const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
         if (t * KBF + row < S) {
-          k_stage[i] = stage_load16(kbase, voff[i], so);
-          v_stage[i] = stage_load16(vbase, voff[i], so);
+          k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
+          v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
         } else {
           k_stage[i] = bf16x8{};
           v_stage[i] = bf16x8{};
@@ -632,6 +652,7 @@ void attn_bwd_dq_kernel(
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
   const __bf16* kbase = k + ibase;
   const __bf16* vbase = v + ibase;
+  const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
   int voff[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -644,16 +665,16 @@ void attn_bwd_dq_kernel(
     if ((t + 1) * KBQ <= S) {  // uniform fast path (cf. fwd)
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
-        k_stage[i] = stage_load16(kbase, voff[i], so);
-        v_stage[i] = stage_load16(vbase, voff[i], so);
+        k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
+        v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
       }
     } else {
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
         const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
         if (t * KBQ + row < S) {
-          k_stage[i] = stage_load16(kbase, voff[i], so);
-          v_stage[i] = stage_load16(vbase, voff[i], so);
+          k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
+          v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
         } else {
           k_stage[i] = bf16x8{};
           v_stage[i] = bf16x8{};
@@ -889,6 +910,8 @@ void attn_bwd_dkdv_kernel(
   // input vs contiguous dout) -> separate descriptors/voffsets.
   const __bf16* qbase = q + ibase;
   const __bf16* obase_p = dout + obase;
+  const long ext_q = ((long)(S - 1) * rs_i + D) * 2;
+  const long ext_o = ((long)(S - 1) * rs_o + D) * 2;
   int voff_q[NCHUNK], voff_o[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -905,16 +928,16 @@ void attn_bwd_dkdv_kernel(
     if ((t + 1) * QTF <= S) {  // uniform fast path (cf. fwd)
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
-        q_stage[i] = stage_load16(qbase, voff_q[i], t * tsoff_q);
-        do_stage[i] = stage_load16(obase_p, voff_o[i], t * tsoff_o);
+        q_stage[i] = stage_load16x(qbase, ext_q, voff_q[i], t * tsoff_q);
+        do_stage[i] = stage_load16x(obase_p, ext_o, voff_o[i], t * tsoff_o);
       }
     } else {
 #pragma unroll
       for (int i = 0; i < NCHUNK; ++i) {
         const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
         if (t * QTF + row < S) {
-          q_stage[i] = stage_load16(qbase, voff_q[i], t * tsoff_q);
-          do_stage[i] = stage_load16(obase_p, voff_o[i], t * tsoff_o);
+          q_stage[i] = stage_load16x(qbase, ext_q, voff_q[i], t * tsoff_q);
+          do_stage[i] = stage_load16x(obase_p, ext_o, voff_o[i], t * tsoff_o);
         } else {
           q_stage[i] = bf16x8{};
           do_stage[i] = bf16x8{};
