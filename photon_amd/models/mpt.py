@@ -25,6 +25,7 @@ import torch.nn.functional as F
 from ..ops.attention import alibi_slopes, flash_attention, flash_attention_qkv
 from ..ops.cross_entropy import fused_cross_entropy
 from ..ops.layernorm import FusedLayerNorm
+from ..ops.linear import lt_available, lt_linear, lt_mlp
 
 
 @dataclass
@@ -83,13 +84,22 @@ class MPTAttention(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, D = x.shape
         H, dh = self.n_heads, self.d_head
-        qkv = self.Wqkv(x)
+        # hipBLASLt epilogue path: bias fused into the GEMM, bias-grad
+        # fused into the backward dW GEMM (BGRADB) — no torch bias kernels
+        fused = lt_available(x)
+        qkv = (
+            lt_linear(x, self.Wqkv.weight, self.Wqkv.bias)
+            if fused
+            else self.Wqkv(x)
+        )
         # Packed path: the HIP kernels read [B,S,3,H,dh] strided directly
         # (no chunk/transpose/contiguous copies); falls back to reshape +
         # SDPA on CPU / attn_impl=torch.
         out = flash_attention_qkv(
             qkv, H, self.slopes, causal=True, impl=self.cfg.attn_impl
         )  # [B, S, H*dh]
+        if fused:
+            return lt_linear(out, self.out_proj.weight, self.out_proj.bias)
         return self.out_proj(out)
 
 
@@ -99,10 +109,16 @@ class MPTMLP(nn.Module):
         bias = not cfg.no_bias
         hidden = cfg.expansion_ratio * cfg.d_model
         self.up_proj = nn.Linear(cfg.d_model, hidden, bias=bias)
-        self.act = nn.GELU(approximate="none")
+        # tanh flavor matches the hipBLASLt GELU epilogue (diff vs exact
+        # GELU is below bf16 resolution; keeps CPU/GPU paths comparable)
+        self.act = nn.GELU(approximate="tanh")
         self.down_proj = nn.Linear(hidden, cfg.d_model, bias=bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if lt_available(x):
+            # one autograd node: GELU_AUX_BIAS fwd, DGELU_BGRAD/BGRADB bwd
+            return lt_mlp(x, self.up_proj.weight, self.up_proj.bias,
+                          self.down_proj.weight, self.down_proj.bias)
         return self.down_proj(self.act(self.up_proj(x)))
 
 

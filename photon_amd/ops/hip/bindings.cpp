@@ -43,6 +43,14 @@ std::vector<torch::Tensor> attn_bwd_launch(torch::Tensor dout, torch::Tensor q,
                                            torch::Tensor slopes,
                                            torch::Tensor o, torch::Tensor lse,
                                            bool causal);
+std::vector<torch::Tensor> lt_linear_fwd(torch::Tensor x, torch::Tensor w,
+                                         c10::optional<torch::Tensor> bias,
+                                         bool gelu);
+std::vector<torch::Tensor> lt_linear_bwd_dx(torch::Tensor dy, torch::Tensor w,
+                                            c10::optional<torch::Tensor> aux_z,
+                                            bool want_bgrad);
+std::vector<torch::Tensor> lt_linear_bwd_dw(torch::Tensor x, torch::Tensor dy,
+                                            bool want_bgrad);
 
 }  // namespace photon_hip
 
@@ -58,6 +66,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale_clip", &multi_tensor_scale_clip);
   m.def("attn_fwd", &attn_fwd_launch, "flash attention forward (ALiBi fused)");
   m.def("attn_bwd", &attn_bwd_launch, "flash attention backward");
+  m.def("lt_linear_fwd", &lt_linear_fwd,
+        "hipblaslt linear fwd (bias/GELU epilogue)");
+  m.def("lt_linear_bwd_dx", &lt_linear_bwd_dx,
+        "hipblaslt linear bwd dx (DGELU/BGRAD epilogue)");
+  m.def("lt_linear_bwd_dw", &lt_linear_bwd_dw,
+        "hipblaslt linear bwd dW (BGRADB epilogue)");
   m.def("attn_fwd_qkv", &attn_fwd_qkv,
         "flash attention forward on packed [B,S,3HD] qkv");
   m.def("attn_bwd_qkv", &attn_bwd_qkv,

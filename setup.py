@@ -32,6 +32,7 @@ sources = [
     str(HIP_DIR / "cross_entropy.hip"),
     str(HIP_DIR / "optim.hip"),
     str(HIP_DIR / "attention.hip"),
+    str(HIP_DIR / "linear_lt.hip"),
     str(HIP_DIR / "debug.hip"),
 ]
 
@@ -45,6 +46,7 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},

@@ -388,3 +388,85 @@ def test_adamw_kernel_master_weights():
                    1 - b1, 1 - b2, [master])
     assert torch.allclose(master, pref, atol=1e-6), (master - pref).abs().max()
     assert torch.equal(pb, master.to(torch.bfloat16))
+
+
+# ---------------------------------------------------------------------------
+# hipBLASLt fused linear / MLP epilogues
+# ---------------------------------------------------------------------------
+
+def test_lt_linear_vs_torch(dev, ext):
+    torch.manual_seed(21)
+    M, K, N = 512, 256, 384
+    x = (torch.randn(M, K, device=dev) * 0.5).to(torch.bfloat16).requires_grad_()
+    w = (torch.randn(N, K, device=dev) * 0.05).to(torch.bfloat16).requires_grad_()
+    b = torch.randn(N, device=dev).to(torch.bfloat16).requires_grad_()
+    from photon_amd.ops.linear import lt_linear
+
+    y = lt_linear(x, w, b)
+    dy = torch.randn_like(y) * 0.1
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_()
+    wr = w.detach().float().requires_grad_()
+    br = b.detach().float().requires_grad_()
+    yr = torch.nn.functional.linear(xr, wr, br)
+    yr.backward(dy.float())
+
+    assert (y.float() - yr).abs().max() < 0.1, "fwd mismatch"
+    assert (x.grad.float() - xr.grad).abs().max() < 0.1
+    assert (w.grad.float() - wr.grad).abs().max() / wr.grad.abs().max() < 0.05
+    assert (b.grad.float() - br.grad).abs().max() / br.grad.abs().max() < 0.05
+
+
+def test_lt_mlp_vs_torch(dev, ext):
+    torch.manual_seed(22)
+    M, K, H = 512, 256, 1024
+    x = (torch.randn(M, K, device=dev) * 0.5).to(torch.bfloat16).requires_grad_()
+    wu = (torch.randn(H, K, device=dev) * 0.05).to(torch.bfloat16).requires_grad_()
+    bu = torch.randn(H, device=dev).mul(0.1).to(torch.bfloat16).requires_grad_()
+    wd = (torch.randn(K, H, device=dev) * 0.05).to(torch.bfloat16).requires_grad_()
+    bd = torch.randn(K, device=dev).mul(0.1).to(torch.bfloat16).requires_grad_()
+    from photon_amd.ops.linear import lt_mlp
+
+    y = lt_mlp(x, wu, bu, wd, bd)
+    dy = torch.randn_like(y) * 0.1
+    y.backward(dy)
+
+    xr = x.detach().float().requires_grad_()
+    wur = wu.detach().float().requires_grad_()
+    bur = bu.detach().float().requires_grad_()
+    wdr = wd.detach().float().requires_grad_()
+    bdr = bd.detach().float().requires_grad_()
+    h = torch.nn.functional.gelu(
+        torch.nn.functional.linear(xr, wur, bur), approximate="tanh"
+    )
+    yr = torch.nn.functional.linear(h, wdr, bdr)
+    yr.backward(dy.float())
+
+    assert (y.float() - yr).abs().max() < 0.15, "mlp fwd mismatch"
+    assert (x.grad.float() - xr.grad).abs().max() < 0.1
+    for got, ref in [(wu.grad, wur.grad), (wd.grad, wdr.grad),
+                     (bu.grad, bur.grad), (bd.grad, bdr.grad)]:
+        rel = (got.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-6)
+        assert rel < 0.06, f"grad rel err {rel}"
+
+
+def test_model_forward_uses_lt_path(dev, ext):
+    """End-to-end: a tiny MPT block fwd+bwd on GPU runs through the fused
+    path (lt_available) and produces finite grads."""
+    from photon_amd.models import build_model
+
+    torch.manual_seed(23)
+    cfg = {
+        "model": {
+            "d_model": 128, "n_heads": 2, "n_layers": 2,
+            "expansion_ratio": 4, "max_seq_len": 128, "vocab_size": 512,
+            "attn_config": {"attn_impl": "flash"},
+        }
+    }
+    m = build_model(cfg).to(dev).to(torch.bfloat16)
+    ids = torch.randint(0, 512, (2, 128), device=dev)
+    out = m(ids, labels=ids)
+    out["loss"].backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n
