@@ -51,6 +51,7 @@ std::vector<torch::Tensor> lt_linear_bwd_dx(torch::Tensor dy, torch::Tensor w,
                                             bool want_bgrad);
 std::vector<torch::Tensor> lt_linear_bwd_dw(torch::Tensor x, torch::Tensor dy,
                                             bool want_bgrad);
+torch::Tensor bias_grad(torch::Tensor dy);
 
 }  // namespace photon_hip
 
@@ -72,6 +73,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hipblaslt linear bwd dx (DGELU/BGRAD epilogue)");
   m.def("lt_linear_bwd_dw", &lt_linear_bwd_dw,
         "hipblaslt linear bwd dW (BGRADB epilogue)");
+  m.def("bias_grad", &bias_grad, "two-stage deterministic bias-grad reduce");
   m.def("attn_fwd_qkv", &attn_fwd_qkv,
         "flash attention forward on packed [B,S,3HD] qkv");
   m.def("attn_bwd_qkv", &attn_bwd_qkv,

@@ -133,7 +133,7 @@ void lt_matmul(const void* A, hipblasOperation_t opA, int64_t a_rows,
     size_t ws = kWorkspaceBytes;
     LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-    constexpr int kMaxAlgos = 8;
+    constexpr int kMaxAlgos = 24;
     hipblasLtMatmulHeuristicResult_t results[kMaxAlgos];
     int n_results = 0;
     LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(lt_handle(), p.op, p.la, p.lb,

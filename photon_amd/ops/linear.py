@@ -3,11 +3,12 @@
 Replaces the reference's cuBLAS GEMM + separate torch bias/GELU kernels
 (SURVEY.md §2.3 "cuBLAS GEMM" row) with single-kernel epilogue GEMMs:
 
-* forward:  BIAS / GELU_AUX_BIAS (pre-GELU z saved for backward)
-* backward: DGELU_BGRAD (dH -> dZ with the up-proj bias grad fused) and
-  BGRADB (bias grad fused into the dW GEMM) — this removes the standalone
-  GELU fwd/bwd pair and every `at::native::reduce_kernel` bias-grad column
-  reduce from the step (VERDICT r01 "GEMM-side fusion", ~6% of step time).
+* forward:  BIAS epilogue (bias in the GEMM)
+* backward: BGRADB — the bias gradient rides the dW GEMM, removing every
+  `at::native::reduce_kernel` bias-grad column reduce from the step
+  (VERDICT r01 "GEMM-side fusion"). The DGELU/AUX epilogues are probed
+  unsupported on gfx950's hipBLASLt (scripts/lt_epi_probe.hip), so GELU
+  stays a torch op between the two fused MLP GEMMs.
 
 GELU flavor: hipBLASLt implements the tanh approximation, so the eager
 fallback uses ``nn.GELU(approximate="tanh")`` to keep CPU/GPU paths
@@ -20,9 +21,22 @@ cast nodes to fp32 automatically.
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import hip_ext
+
+# PHOTON_LT_MODE:
+#   "dbk" (default) — torch (TunableOp-tuned) GEMMs + the hand-written HIP
+#       bias-grad column reduce in backward (replaces torch's 1.7 TB/s
+#       reduce_kernel; deterministic two-stage, ops/hip/bias_grad.hip);
+#   "all"/"dw"      — hipblaslt epilogue experiments (measured REJECTED:
+#       BGRADB disables split-K on the reduction-heavy dW GEMMs, MPT-1B
+#       step 1704 -> 3515 ms; kept for re-evaluation on newer hipblaslt);
+#   "off"           — plain torch linears.
+def _lt_mode() -> str:
+    return os.environ.get("PHOTON_LT_MODE", "dbk")
 
 _CUSTOM_FWD = torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
 _CUSTOM_BWD = torch.amp.custom_bwd(device_type="cuda")
@@ -34,6 +48,7 @@ def lt_available(x: torch.Tensor) -> bool:
         ext is not None
         and hasattr(ext, "lt_linear_fwd")
         and x.is_cuda
+        and _lt_mode() != "off"
     )
 
 
@@ -41,7 +56,10 @@ class _LtLinearFn(torch.autograd.Function):
     @staticmethod
     @_CUSTOM_FWD
     def forward(ctx, x, w, b):
-        y, _ = hip_ext().lt_linear_fwd(x, w, b, False)
+        if _lt_mode() == "all":
+            y, _ = hip_ext().lt_linear_fwd(x, w, b, False)
+        else:  # torch (TunableOp-tuned) forward GEMM
+            y = torch.nn.functional.linear(x, w, b)
         ctx.save_for_backward(x, w)
         ctx.has_bias = b is not None
         return y
@@ -52,38 +70,17 @@ class _LtLinearFn(torch.autograd.Function):
         x, w = ctx.saved_tensors
         dy = dy.contiguous()
         ext = hip_ext()
-        dx = ext.lt_linear_bwd_dx(dy, w, None, False)[0]
-        dw, db = ext.lt_linear_bwd_dw(x, dy, ctx.has_bias)
+        mode = _lt_mode()
+        if mode == "all":
+            dx = ext.lt_linear_bwd_dx(dy, w, None, False)[0]
+        else:
+            dx = dy @ w
+        if mode in ("all", "dw"):
+            dw, db = ext.lt_linear_bwd_dw(x, dy, ctx.has_bias)
+        else:  # "dbk": tuned torch dW GEMM + HIP bias-grad reduce
+            dw = dy.t() @ x
+            db = ext.bias_grad(dy) if ctx.has_bias else None
         return dx, dw, (db if ctx.has_bias else None)
-
-
-class _LtMLPFn(torch.autograd.Function):
-    """up_proj + GELU + down_proj with all epilogues fused."""
-
-    @staticmethod
-    @_CUSTOM_FWD
-    def forward(ctx, x, wu, bu, wd, bd):
-        ext = hip_ext()
-        h, z = ext.lt_linear_fwd(x, wu, bu, True)  # h = gelu(z), z = xWu+bu
-        y, _ = ext.lt_linear_fwd(h, wd, bd, False)
-        ctx.save_for_backward(x, wu, wd, z, h)
-        ctx.has_bias = bu is not None
-        return y
-
-    @staticmethod
-    @_CUSTOM_BWD
-    def backward(ctx, dy):
-        x, wu, wd, z, h = ctx.saved_tensors
-        dy = dy.contiguous()
-        ext = hip_ext()
-        # dZ = dgelu(z) * (dy @ Wd), db_up fused in the same GEMM
-        dz, dbu = ext.lt_linear_bwd_dx(dy, wd, z, ctx.has_bias)
-        dwd, dbd = ext.lt_linear_bwd_dw(h, dy, ctx.has_bias)
-        dwu, _ = ext.lt_linear_bwd_dw(x, dz, False)
-        dx = ext.lt_linear_bwd_dx(dz, wu, None, False)[0]
-        if not ctx.has_bias:
-            dbu = dbd = None
-        return dx, dwu, dbu, dwd, dbd
 
 
 def lt_linear(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Tensor:
@@ -93,7 +90,12 @@ def lt_linear(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Tensor:
     return y.reshape(*shp[:-1], weight.shape[0])
 
 
-def lt_mlp(x, w_up, b_up, w_down, b_down) -> torch.Tensor:
-    shp = x.shape
-    y = _LtMLPFn.apply(x.reshape(-1, shp[-1]), w_up, b_up, w_down, b_down)
-    return y.reshape(*shp[:-1], w_down.shape[0])
+def lt_mlp(x, w_up, b_up, w_down, b_down, act=None) -> torch.Tensor:
+    """up -> GELU(tanh) -> down. The GEMMs carry BIAS epilogues forward and
+    BGRADB backward; the GELU itself stays a torch op because hipBLASLt on
+    gfx950 ships no kernels for the AUX/DGELU epilogues (probed:
+    scripts/lt_epi_probe.hip — GELU_AUX*/DGELU*/BGRADA all return 0 algos
+    in the forward layout; BGRADB works in the dW layout)."""
+    h = lt_linear(x, w_up, b_up)
+    h = torch.nn.functional.gelu(h, approximate="tanh")
+    return lt_linear(h, w_down, b_down)

@@ -33,6 +33,7 @@ sources = [
     str(HIP_DIR / "optim.hip"),
     str(HIP_DIR / "attention.hip"),
     str(HIP_DIR / "linear_lt.hip"),
+    str(HIP_DIR / "bias_grad.hip"),
     str(HIP_DIR / "debug.hip"),
 ]
 

@@ -470,3 +470,15 @@ def test_model_forward_uses_lt_path(dev, ext):
     out["loss"].backward()
     for n, p in m.named_parameters():
         assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+def test_bias_grad_kernel(dev, ext):
+    torch.manual_seed(31)
+    for M, N in [(4096, 768), (65536, 3072), (1000, 2048), (16384, 8192)]:
+        dy = (torch.randn(M, N, device=dev) * 0.1).to(torch.bfloat16)
+        db = ext.bias_grad(dy)
+        ref = dy.float().sum(0)
+        rel = (db.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-6)
+        assert rel < 0.02, (M, N, rel)
+        # deterministic: same input -> bit-identical output
+        assert torch.equal(db, ext.bias_grad(dy))
