@@ -52,6 +52,8 @@ std::vector<torch::Tensor> lt_linear_bwd_dx(torch::Tensor dy, torch::Tensor w,
 std::vector<torch::Tensor> lt_linear_bwd_dw(torch::Tensor x, torch::Tensor dy,
                                             bool want_bgrad);
 torch::Tensor bias_grad(torch::Tensor dy);
+torch::Tensor fused_linear(torch::Tensor x, torch::Tensor w,
+                           c10::optional<torch::Tensor> b);
 
 }  // namespace photon_hip
 
@@ -74,6 +76,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lt_linear_bwd_dw", &lt_linear_bwd_dw,
         "hipblaslt linear bwd dW (BGRADB epilogue)");
   m.def("bias_grad", &bias_grad, "two-stage deterministic bias-grad reduce");
+  m.def("fused_linear", &fused_linear,
+        "C++ autograd linear: tuned GEMMs + HIP bias-grad backward");
   m.def("attn_fwd_qkv", &attn_fwd_qkv,
         "flash attention forward on packed [B,S,3HD] qkv");
   m.def("attn_bwd_qkv", &attn_bwd_qkv,

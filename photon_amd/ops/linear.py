@@ -28,15 +28,18 @@ import torch
 from . import hip_ext
 
 # PHOTON_LT_MODE:
-#   "dbk" (default) — torch (TunableOp-tuned) GEMMs + the hand-written HIP
-#       bias-grad column reduce in backward (replaces torch's 1.7 TB/s
-#       reduce_kernel; deterministic two-stage, ops/hip/bias_grad.hip);
+#   "cpp" (default) — C++ autograd Function (zero Python overhead): torch
+#       tuned GEMMs + the HIP bias-grad column reduce in backward
+#       (ops/hip/fused_linear_fn.cpp + bias_grad.hip);
+#   "dbk"           — same structure as a Python autograd.Function
+#       (measured: ~30 ms/step of Python Function overhead at 125M —
+#       kept only as the measurement record);
 #   "all"/"dw"      — hipblaslt epilogue experiments (measured REJECTED:
 #       BGRADB disables split-K on the reduction-heavy dW GEMMs, MPT-1B
 #       step 1704 -> 3515 ms; kept for re-evaluation on newer hipblaslt);
 #   "off"           — plain torch linears.
 def _lt_mode() -> str:
-    return os.environ.get("PHOTON_LT_MODE", "dbk")
+    return os.environ.get("PHOTON_LT_MODE", "cpp")
 
 _CUSTOM_FWD = torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
 _CUSTOM_BWD = torch.amp.custom_bwd(device_type="cuda")
@@ -84,7 +87,9 @@ class _LtLinearFn(torch.autograd.Function):
 
 
 def lt_linear(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Tensor:
-    """y = x @ W^T + b via hipBLASLt (3-D x flattened to 2-D)."""
+    """y = x @ W^T + b with the HIP bias-grad backward."""
+    if _lt_mode() == "cpp":
+        return hip_ext().fused_linear(x, weight, bias)
     shp = x.shape
     y = _LtLinearFn.apply(x.reshape(-1, shp[-1]), weight, bias)
     return y.reshape(*shp[:-1], weight.shape[0])

@@ -34,6 +34,7 @@ sources = [
     str(HIP_DIR / "attention.hip"),
     str(HIP_DIR / "linear_lt.hip"),
     str(HIP_DIR / "bias_grad.hip"),
+    str(HIP_DIR / "fused_linear_fn.cpp"),
     str(HIP_DIR / "debug.hip"),
 ]
 

@@ -482,3 +482,35 @@ def test_bias_grad_kernel(dev, ext):
         assert rel < 0.02, (M, N, rel)
         # deterministic: same input -> bit-identical output
         assert torch.equal(db, ext.bias_grad(dy))
+
+
+def test_cpp_fused_linear_grads(dev, ext):
+    """C++ autograd fused_linear vs torch F.linear: values and all grads,
+    both direct bf16 and under autocast from fp32 masters."""
+    torch.manual_seed(41)
+    M, K, N = 256, 128, 192
+    for autocast_mode in (False, True):
+        dt = torch.float32 if autocast_mode else torch.bfloat16
+        x = (torch.randn(M, K, device=dev) * 0.5).to(dt).requires_grad_()
+        w = (torch.randn(N, K, device=dev) * 0.05).to(dt).requires_grad_()
+        b = torch.randn(N, device=dev).to(dt).requires_grad_()
+        xr = x.detach().clone().requires_grad_()
+        wr = w.detach().clone().requires_grad_()
+        br = b.detach().clone().requires_grad_()
+        ctx = (
+            torch.autocast("cuda", dtype=torch.bfloat16)
+            if autocast_mode
+            else torch.autocast("cuda", enabled=False)
+        )
+        with ctx:
+            y = ext.fused_linear(x, w, b)
+            yr = torch.nn.functional.linear(xr, wr, br)
+        dy = torch.randn_like(y.float()) * 0.1
+        y.backward(dy.to(y.dtype))
+        yr.backward(dy.to(yr.dtype))
+        assert (y.float() - yr.float()).abs().max() < 1e-3
+        assert (x.grad - xr.grad).abs().max() < 1e-2, autocast_mode
+        relw = (w.grad - wr.grad).abs().max() / wr.grad.abs().max()
+        assert relw < 0.03, (autocast_mode, relw)
+        relb = (b.grad - br.grad).abs().max() / br.grad.abs().max()
+        assert relb < 0.03, (autocast_mode, relb)
