@@ -6,7 +6,7 @@ cd /tmp
 for MODE in "$@"; do
   PHOTON_LT_MODE=$MODE rocprofv3 --kernel-trace --stats --output-format csv \
     -d /root/repo/gpurun_out/prof_$MODE -o st -- \
-    /bin/bash -c "timeout 240 python bench.py --gpus 1 --steps 4 --warmup 2 > /dev/null 2>&1" \
+    /bin/bash -c "cd /root/repo && timeout 240 python bench.py --gpus 1 --steps 4 --warmup 2 > /dev/null 2>&1" \
     > /tmp/prof_$MODE.log 2>&1 || { echo "prof $MODE failed"; tail -3 /tmp/prof_$MODE.log; }
 done
 cd /root/repo
