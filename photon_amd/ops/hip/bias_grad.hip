@@ -1,19 +1,19 @@
 // Bias-gradient column reduce: db[n] = sum_m dy[m, n] — replaces
-// at::native::reduce_kernel (measured 1.7 TB/s, ~2.7% of the 125M step)
-// with a coalesced two-stage reduction.
+// at::native::reduce_kernel (~2.7% of the 125M step at 1.7 TB/s) with a
+// two-stage deterministic reduction.
 //
-// Stage 1: grid (m_chunks, ceil(N/1024)); each 256-thread block owns 1024
-// consecutive columns (4 bf16 = one 8-byte load per thread per row) and a
-// contiguous row range, accumulating fp32 partials — consecutive threads
-// read consecutive columns, so each row access is one 2 KB contiguous
-// burst per block. Stage 2 folds the m_chunks partials. No atomics:
-// bit-deterministic (fixed reduction order), matching the attention
-// backward's atomics-free design.
+// v1 lesson (measured): an un-unrolled row loop is LATENCY-bound (~1 TB/s
+// — one 8-byte load in flight per lane) and a 3-block stage 2 with serial
+// strided loads cost 60 us/call. v2: stage 1 unrolls 8 rows (8 loads in
+// flight), the chunk count scales with N so the grid fills all 256 CUs
+// (~2048 blocks), and stage 2 parallelizes over chunks WITHIN a block
+// (LDS tree) instead of one serial chain per column.
+//
+// No atomics: fixed reduction order, bit-deterministic.
 //
 // hipBLASLt's BGRADB epilogue was measured as the alternative and rejected:
-// fusing the bias grad into the dW GEMM disables split-K kernel selection,
-// and the reduction-heavy dW shapes (k = tokens) then run ~10x slower
-// (MPT-1B step 1704 -> 3515 ms). See scripts/lt_epi_probe.hip notes.
+// fusing the bias grad into the dW GEMM disables split-K selection and the
+// reduction-heavy dW shapes run ~10x slower (MPT-1B 1704 -> 3515 ms/step).
 
 #include "host_common.h"
 
@@ -30,7 +30,23 @@ __global__ __launch_bounds__(256) void bias_grad_stage1(
   const long m1 = min(m0 + rows_per_chunk, M);
   float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
   const __bf16* p = dy + m0 * N + col0;
-  for (long m = m0; m < m1; ++m, p += N) {
+  long m = m0;
+  // 8 rows in flight per lane: the loads are independent, the adds chain
+  // per accumulator only
+  for (; m + 8 <= m1; m += 8) {
+    bf16x4g v[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v[u] = *(const bf16x4g*)(p + u * N);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      acc0 += (float)v[u][0];
+      acc1 += (float)v[u][1];
+      acc2 += (float)v[u][2];
+      acc3 += (float)v[u][3];
+    }
+    p += 8 * N;
+  }
+  for (; m < m1; ++m, p += N) {
     const bf16x4g v = *(const bf16x4g*)p;
     acc0 += (float)v[0];
     acc1 += (float)v[1];
@@ -44,15 +60,30 @@ __global__ __launch_bounds__(256) void bias_grad_stage1(
   out[3] = acc3;
 }
 
+// One block per 4 columns; 256 threads split as 64 chunk-lanes x 4 columns.
+// Each lane strides the chunk dimension (independent loads), then an LDS
+// tree folds the 64 partial sums per column in a fixed order.
 template <typename OutT>
 __global__ __launch_bounds__(256) void bias_grad_stage2(
     const float* __restrict__ partial, OutT* __restrict__ db, long N,
     int m_chunks) {
-  const long col = (long)blockIdx.x * 256 + threadIdx.x;
-  if (col >= N) return;
+  __shared__ float red[256];
+  const int cl = threadIdx.x & 3;        // column within the quad
+  const int lane = threadIdx.x >> 2;     // chunk lane 0..63
+  const long col = (long)blockIdx.x * 4 + cl;
   float acc = 0.f;
-  for (int c = 0; c < m_chunks; ++c) acc += partial[(long)c * N + col];
-  db[col] = (OutT)acc;
+  if (col < N) {
+    for (int c = lane; c < m_chunks; c += 64)
+      acc += partial[(long)c * N + col];
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+#pragma unroll
+  for (int step = 128; step >= 4; step >>= 1) {
+    if (threadIdx.x < step) red[threadIdx.x] += red[threadIdx.x + step];
+    __syncthreads();
+  }
+  if (threadIdx.x < 4 && col < N) db[col] = (OutT)red[threadIdx.x];
 }
 
 // dy: [M, N] bf16 contiguous -> returns db [N] in dy's dtype.
@@ -62,11 +93,10 @@ torch::Tensor bias_grad(torch::Tensor dy) {
               "bias_grad: need contiguous 2-D bf16");
   const long M = dy.size(0), N = dy.size(1);
   TORCH_CHECK(N % 4 == 0, "bias_grad: N must be a multiple of 4");
-  // enough chunks to fill the chip (>=1024 blocks total), capped so the
-  // partial buffer stays tiny
+  // fill the chip: ~2048 blocks regardless of N
   const long col_blocks = (N + 1023) / 1024;
-  int m_chunks = (int)std::min<long>((1024 + col_blocks - 1) / col_blocks,
-                                     (M + 255) / 256);
+  int m_chunks = (int)std::min<long>((2048 + col_blocks - 1) / col_blocks,
+                                     (M + 31) / 32);
   m_chunks = std::max(m_chunks, 1);
   const long rows_per_chunk = (M + m_chunks - 1) / m_chunks;
   auto partial = at::empty({m_chunks, N}, dy.options().dtype(at::kFloat));
@@ -75,7 +105,7 @@ torch::Tensor bias_grad(torch::Tensor dy) {
   hipLaunchKernelGGL(bias_grad_stage1, g1, dim3(256), 0, cur_stream(),
                      (const __bf16*)dy.data_ptr(), partial.data_ptr<float>(),
                      M, N, rows_per_chunk);
-  hipLaunchKernelGGL((bias_grad_stage2<__bf16>), dim3((N + 255) / 256),
+  hipLaunchKernelGGL((bias_grad_stage2<__bf16>), dim3((N + 3) / 4),
                      dim3(256), 0, cur_stream(), partial.data_ptr<float>(),
                      (__bf16*)db.data_ptr(), N, m_chunks);
   return db;
