@@ -46,6 +46,9 @@ class ByteTokenizer:
     def encode(self, text: str) -> list[int]:
         return list(text.encode("utf-8"))
 
+    def decode(self, ids) -> str:
+        return bytes(int(i) & 0xFF for i in ids).decode("utf-8", "ignore")
+
     def save_pretrained(self, path):
         Path(path).mkdir(parents=True, exist_ok=True)
         (Path(path) / "tokenizer_config.json").write_text(

@@ -1,25 +1,33 @@
-"""ICL task evaluation + Eval Gauntlet composite.
+"""ICL task evaluation + Eval Gauntlet composite (reference v0.3 surface).
 
 Re-implements the slice of llm-foundry's ICL harness the reference drives
 through ``icl_tasks_config`` / ``eval_gauntlet_config``
-(conf/{icl_tasks,eval_gauntlet}_config/*; both default to
-``empty`` and are config-gated — photon/conf/base.yaml defaults block).
+(conf/{icl_tasks,eval_gauntlet}_config/*). Consumes the
+reference's ``tasks_v0.3.yaml`` structure (all four task types +
+num_fewshot lists, question_prelimiter, cot_delimiter,
+early_stopping_criteria, do_normalization — reference
+photon/conf/icl_tasks_config/tasks_v0.3.yaml) and the
+``eval_gauntlet_v0.3.yaml`` composite semantics (EQUAL weighting,
+subtract_random_baseline, rescale_accuracy, named averages — reference
+photon/conf/eval_gauntlet_config/eval_gauntlet_v0.3.yaml:1).
 
-Task format: jsonl files with {"context": ..., "continuation": ...} for
-``language_modeling`` tasks, or {"query": ..., "choices": [...],
-"gold": idx} for ``multiple_choice`` — the same fields llm-foundry's ICL
-datasets carry. Scoring:
+Task jsonl formats (llm-foundry ICL dataset fields):
 
-* language_modeling  -> per-token greedy accuracy over the continuation
-* multiple_choice    -> accuracy of argmin over length-normalized CE
+* language_modeling              {"context", "continuation"}
+* multiple_choice                {"query", "choices", "gold"}
+* schema                         {"context_options", "continuation", "gold"}
+* generation_task_with_answers   {"context", "answer", "aliases"}
 
-The gauntlet composite is the category-weighted mean of task scores
-(llm-foundry eval_gauntlet semantics).
+Scoring: per-token greedy accuracy (LM), argmin length-normalized CE (MC),
+argmin continuation CE over context options (schema), normalized exact
+match of a greedy generation (QA).
 """
 
 from __future__ import annotations
 
 import json
+import re
+import string
 from pathlib import Path
 
 import torch
@@ -28,7 +36,7 @@ import torch.nn.functional as F
 
 def _resolve_task_path(path: str | Path) -> Path:
     """Relative task paths resolve against the photon_amd package dir
-    (bundled demo tasks), falling back to the cwd."""
+    (bundled demo/local_data tasks), falling back to the cwd."""
     p = Path(path)
     if not p.is_absolute() and not p.exists():
         pkg = Path(__file__).resolve().parent.parent / p
@@ -47,6 +55,15 @@ def load_jsonl_task(path: str | Path) -> list[dict]:
     return out
 
 
+def _normalize_answer(s: str) -> str:
+    """SQuAD-style normalization (llm-foundry do_normalization): lowercase,
+    strip punctuation/articles/extra whitespace."""
+    s = s.lower()
+    s = "".join(ch for ch in s if ch not in set(string.punctuation))
+    s = re.sub(r"\b(a|an|the)\b", " ", s)
+    return " ".join(s.split())
+
+
 @torch.no_grad()
 def _continuation_stats(model, tok, context: str, continuation: str, device,
                         max_len: int):
@@ -58,7 +75,6 @@ def _continuation_stats(model, tok, context: str, continuation: str, device,
         return 0.0, 0, 0.0
     x = torch.tensor([ids], dtype=torch.long, device=device)
     logits = model(x)["logits"][0]  # [T, V]
-    # positions predicting the continuation tokens
     tgt = x[0, -n_cont:]
     pred_logits = logits[-n_cont - 1 : -1]
     ce = float(F.cross_entropy(pred_logits.float(), tgt, reduction="sum"))
@@ -67,24 +83,84 @@ def _continuation_stats(model, tok, context: str, continuation: str, device,
 
 
 @torch.no_grad()
+def _greedy_generate(model, tok, prompt: str, device, max_len: int,
+                     max_new_tokens: int = 24,
+                     stop_strings: list[str] | None = None) -> str:
+    ids = tok.encode(prompt)[-(max_len - max_new_tokens):]
+    x = torch.tensor([ids], dtype=torch.long, device=device)
+    out_ids: list[int] = []
+    for _ in range(max_new_tokens):
+        logits = model(x)["logits"][0, -1]
+        nxt = int(logits.argmax())
+        out_ids.append(nxt)
+        x = torch.cat([x, torch.tensor([[nxt]], device=device)], dim=1)
+        text = tok.decode(out_ids)
+        for stop in stop_strings or []:
+            if stop in text:
+                return text.split(stop)[0]
+    return tok.decode(out_ids)
+
+
+def _fewshot_prefix(examples: list[dict], i: int, k: int, kind: str,
+                    delim: str, prelim: str) -> str:
+    """k in-context examples drawn deterministically from the OTHER
+    examples (llm-foundry uses a seeded sample; index-skip is equivalent
+    determinism for the bundled local data)."""
+    if k <= 0:
+        return ""
+    parts = []
+    j = 0
+    while len(parts) < k and j < len(examples):
+        if j != i:
+            ex = examples[j]
+            if kind == "multiple_choice":
+                parts.append(prelim + ex["query"] + delim +
+                             ex["choices"][int(ex["gold"])])
+            elif kind == "schema":
+                parts.append(ex["context_options"][int(ex["gold"])] + delim +
+                             ex["continuation"])
+            elif kind == "generation_task_with_answers":
+                parts.append(prelim + ex["context"] + delim + str(ex["answer"]))
+            else:
+                parts.append(prelim + ex["context"] + delim +
+                             ex["continuation"])
+        j += 1
+    return "\n".join(parts) + "\n" if parts else ""
+
+
+@torch.no_grad()
 def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                        max_seq_len: int = 2048,
                        limit_examples: int | None = None) -> dict[str, float]:
-    """Returns {"metrics/icl/{label}/accuracy": v, ...} per task."""
+    """Returns {"metrics/icl/{label}/accuracy": v, ...} per task; tasks
+    whose dataset_uri is absent are skipped with a metric of nan (the
+    reference's remote datasets are not fetchable offline)."""
     model.eval()
     results: dict[str, float] = {}
     for task in tasks_cfg or []:
         label = str(task.get("label", "task"))
         kind = str(task.get("icl_task_type", "language_modeling"))
         delim = str(task.get("continuation_delimiter", " "))
-        examples = load_jsonl_task(task["dataset_uri"])
+        prelim = str(task.get("question_prelimiter", ""))
+        cot = task.get("cot_delimiter")
+        stops = list(task.get("early_stopping_criteria") or [])
+        nf = task.get("num_fewshot", [0])
+        n_fewshot = int(nf[0]) if isinstance(nf, (list, tuple)) else int(nf)
+        do_norm = bool(task.get("do_normalization", True))
+        try:
+            examples = load_jsonl_task(task["dataset_uri"])
+        except OSError:
+            results[f"metrics/icl/{label}/accuracy"] = float("nan")
+            continue
         if limit_examples:
             examples = examples[:limit_examples]
         correct, total = 0, 0
-        for ex in examples:
+        for i, ex in enumerate(examples):
+            shots = _fewshot_prefix(examples, i, n_fewshot, kind, delim,
+                                    prelim)
             if kind == "language_modeling":
                 ce, n, c = _continuation_stats(
-                    model, tokenizer, ex["context"] + delim,
+                    model, tokenizer, shots + ex["context"] + delim,
                     ex["continuation"], device, max_seq_len,
                 )
                 correct += c
@@ -93,12 +169,42 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                 ces = []
                 for choice in ex["choices"]:
                     ce, n, _ = _continuation_stats(
-                        model, tokenizer, ex["query"] + delim, choice,
-                        device, max_seq_len,
+                        model, tokenizer, shots + prelim + ex["query"] + delim,
+                        choice, device, max_seq_len,
                     )
                     ces.append(ce / max(n, 1))
                 correct += int(min(range(len(ces)), key=ces.__getitem__)
                                == int(ex["gold"]))
+                total += 1
+            elif kind == "schema":
+                # choose the context option that best explains the SAME
+                # continuation (winograd-style)
+                ces = []
+                for opt in ex["context_options"]:
+                    ce, n, _ = _continuation_stats(
+                        model, tokenizer, shots + opt + delim,
+                        ex["continuation"], device, max_seq_len,
+                    )
+                    ces.append(ce / max(n, 1))
+                correct += int(min(range(len(ces)), key=ces.__getitem__)
+                               == int(ex["gold"]))
+                total += 1
+            elif kind == "generation_task_with_answers":
+                gen = _greedy_generate(
+                    model, tokenizer, shots + prelim + ex["context"] + delim,
+                    device, max_seq_len, stop_strings=stops,
+                )
+                if cot and cot in gen:
+                    gen = gen.split(cot)[-1]
+                answers = [str(ex.get("answer", ""))] + [
+                    str(a) for a in ex.get("aliases", [])
+                ]
+                if do_norm:
+                    gen_n = _normalize_answer(gen)
+                    ok = any(_normalize_answer(a) == gen_n for a in answers)
+                else:
+                    ok = any(a.strip() == gen.strip() for a in answers)
+                correct += int(ok)
                 total += 1
             else:
                 raise ValueError(f"unknown icl_task_type {kind!r}")
@@ -111,25 +217,46 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
 
 def gauntlet_composite(task_results: dict[str, float],
                        gauntlet_cfg: dict | None) -> dict[str, float]:
-    """Category-weighted composite (eval_gauntlet semantics): categories are
-    {name, benchmarks: [{name, weight}]}; composite = mean over categories
-    of the weighted benchmark mean."""
+    """Eval-gauntlet v0.3 composite (reference eval_gauntlet_v0.3.yaml):
+    per-benchmark accuracy, optionally baseline-subtracted and rescaled
+    (acc' = (acc - rb) / (1 - rb)), EQUAL-weighted (or per-benchmark
+    ``weight``) within each category; named ``averages`` over category
+    scores plus an overall ``average``."""
     if not gauntlet_cfg:
         return {}
-    out = {}
-    cat_scores = []
+    import math
+
+    subtract = bool(gauntlet_cfg.get("subtract_random_baseline", False))
+    rescale = bool(gauntlet_cfg.get("rescale_accuracy", False))
+    out: dict[str, float] = {}
+    cat_scores: dict[str, float] = {}
     for cat in gauntlet_cfg.get("categories", []):
         num, den = 0.0, 0.0
         for b in cat.get("benchmarks", []):
             key = f"metrics/icl/{b['name']}/accuracy"
-            if key in task_results:
-                w = float(b.get("weight", 1.0))
-                num += w * task_results[key]
-                den += w
+            if key not in task_results:
+                continue
+            acc = task_results[key]
+            if isinstance(acc, float) and math.isnan(acc):
+                continue
+            rb = float(b.get("random_baseline", 0.0))
+            if subtract:
+                acc = acc - rb
+                if rescale and rb < 1.0:
+                    acc = acc / (1.0 - rb)
+            w = float(b.get("weight", 1.0))
+            num += w * acc
+            den += w
         if den > 0:
             score = num / den
             out[f"metrics/eval_gauntlet/{cat['name']}"] = score
-            cat_scores.append(score)
+            cat_scores[str(cat["name"])] = score
+    for avg_name, members in (gauntlet_cfg.get("averages") or {}).items():
+        vals = [cat_scores[m] for m in members if m in cat_scores]
+        if vals:
+            out[f"metrics/eval_gauntlet/{avg_name}"] = sum(vals) / len(vals)
     if cat_scores:
-        out["metrics/eval_gauntlet/average"] = sum(cat_scores) / len(cat_scores)
+        out["metrics/eval_gauntlet/average"] = (
+            sum(cat_scores.values()) / len(cat_scores)
+        )
     return out
