@@ -57,12 +57,46 @@ class ByteTokenizer:
         )
 
 
-def load_tokenizer(spec: str | None):
+def load_tokenizer(spec: str | None, bos_eos_workaround: bool = True):
     if spec is None:
         return ByteTokenizer()
     from transformers import AutoTokenizer
 
-    return AutoTokenizer.from_pretrained(spec)
+    tok = AutoTokenizer.from_pretrained(spec)
+    if bos_eos_workaround:
+        apply_bos_eos_workaround(tok)
+    return tok
+
+
+def apply_bos_eos_workaround(tok) -> None:
+    """gpt-neox/GPT2-family tokenizers do not insert BOS/EOS even with
+    add_special_tokens — install a TemplateProcessing post-processor
+    (reference photon/dataset/utils.py:66-102). No-op for tokenizers that
+    already insert them or have no bos/eos ids."""
+    bos, eos = getattr(tok, "bos_token_id", None), getattr(tok, "eos_token_id", None)
+    if bos is None or eos is None or not hasattr(tok, "_tokenizer"):
+        return
+    test = tok("test")["input_ids"]
+    if test and (test[0] == bos or test[-1] == eos):
+        return
+    try:
+        from tokenizers.processors import TemplateProcessing
+    except ImportError:
+        return
+    tok._tokenizer.post_processor = TemplateProcessing(
+        single=tok.bos_token + " $A " + tok.eos_token,
+        # no special token between $A and $B: the concat writer already
+        # EOS-joins distinct sequences
+        pair=tok.bos_token + " $A $B " + tok.eos_token,
+        special_tokens=[(tok.eos_token, eos), (tok.bos_token, bos)],
+    )
+    test = tok("test")["input_ids"]
+    if not (test and (test[0] == bos or test[-1] == eos)):
+        raise ValueError(
+            "tokenizer inserts neither BOS nor EOS even after the "
+            "TemplateProcessing workaround; concatenation would attach "
+            "sequences without separators"
+        )
 
 
 def iter_documents(source: str, seed: int = 1337):
@@ -88,6 +122,32 @@ def iter_documents(source: str, seed: int = 1337):
         if hasattr(ds, "values"):  # DatasetDict: use train split
             ds = ds.get("train") or next(iter(ds.values()))
         for row in ds:
+            yield row.get("text", "")
+    elif kind == "hf_stream":
+        # streaming-HF ingestion (reference build_hf_dataset streaming
+        # path): hf_stream:path[:name[:split[:truncate]]] — e.g.
+        # hf_stream:allenai/c4:en:train_small applies the reference split
+        # table truncation (data/constants.py).
+        import datasets
+
+        from .constants import split_spec
+
+        parts = arg.split(":")
+        path = parts[0]
+        name = parts[1] if len(parts) > 1 and parts[1] else None
+        split = parts[2] if len(parts) > 2 else "train"
+        truncate = None
+        try:
+            spec = split_spec(name or "en", split)
+            split_hf, truncate = spec.hf_split, spec.truncated_samples
+        except ValueError:
+            split_hf = split
+        if len(parts) > 3:
+            truncate = int(parts[3])
+        ds = datasets.load_dataset(path, name, split=split_hf, streaming=True)
+        for i, row in enumerate(ds):
+            if truncate is not None and i >= truncate:
+                break
             yield row.get("text", "")
     elif kind == "synthetic":
         rng = np.random.default_rng(seed)

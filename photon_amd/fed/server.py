@@ -188,11 +188,15 @@ class FedServer:
             # final_parameters.npz (init_utils.py:43-125).
             cent_uuid = photon_cfg.get("restore_cent_run_uuid")
             if cent_uuid:
-                npz = (
-                    Path(self.saving_path) / str(cent_uuid)
-                    / "final_parameters.npz"
+                # Composer-format .pt (ep{e}-ba{b}-rank0.pt) or the run's
+                # final_parameters.npz (reference init_utils.py:43-125)
+                from .server_ckpt import get_centralized_run_parameters
+
+                flat = get_centralized_run_parameters(
+                    self.saving_path, str(cent_uuid), self.layout,
+                    photon_cfg.get("restore_cent_run_batches"),
                 )
-                self.layout.from_ndarrays(self.layout.load_npz(npz))
+                self.layout.flat.copy_(flat)
             else:
                 self.layout.copy_from_model(self.client.model)
             self.comm.broadcast_flat(self.layout.flat, src=0)

@@ -183,3 +183,57 @@ def copy_old_checkpoints_to_new_run(
         for cdir in old_base.glob("client_*"):
             shutil.copytree(cdir, new_base / cdir.name, dirs_exist_ok=True)
     return latest
+
+
+def get_centralized_run_parameters(
+    saving_path, cent_run_uuid: str, layout: FlatParams,
+    desired_batches: int | None = None,
+) -> torch.Tensor:
+    """Bootstrap federated params from a CENTRALIZED run's Composer-format
+    checkpoint (reference get_centralized_run_parameters,
+    photon/server/init_utils.py:43-125): enumerate ep{e}-ba{b}-rank0.pt
+    files under the run dir, pick the one whose batch count matches
+    ``photon.restore_cent_run_batches`` (or the latest when unset), load the
+    model state dict ignoring optimizer/scheduler/dataset state, and return
+    the flat fp32 buffer in the wire order. Falls back to the run's
+    final_parameters.npz when no .pt checkpoint exists."""
+    import re
+
+    run_dir = Path(saving_path) / str(cent_run_uuid)
+    pairs = []
+    for p in run_dir.rglob("ep*-ba*-rank0.pt"):
+        m = re.search(r"ep(\d+)-ba(\d+)-rank0\.pt$", p.name)
+        if m:
+            pairs.append((int(m.group(2)), int(m.group(1)), p))
+    if not pairs:
+        npz = run_dir / "final_parameters.npz"
+        if npz.exists():
+            arrays = layout.load_npz(npz)
+            flat = torch.cat([
+                torch.from_numpy(np.ascontiguousarray(a, dtype=np.float32))
+                .reshape(-1) for a in arrays
+            ]).to(layout.flat.device)
+            return flat
+        raise FileNotFoundError(
+            f"no Composer checkpoint or final_parameters.npz under {run_dir}"
+        )
+    pairs.sort()
+    if desired_batches is not None:
+        match = [p for p in pairs if p[0] == int(desired_batches)]
+        if not match:
+            raise ValueError(
+                f"no checkpoint with {desired_batches} batches in {run_dir} "
+                f"(have {[b for b, _, _ in pairs]})"
+            )
+        _, _, path = match[0]
+    else:
+        _, _, path = pairs[-1]
+    ckpt = torch.load(path, map_location="cpu", weights_only=False)
+    model_sd = ckpt["state"]["model"]
+    flat = torch.zeros_like(layout.flat)
+    views = layout.layer_views_of(flat)
+    for name, view in zip(layout.names, views):
+        if name not in model_sd:
+            raise KeyError(f"parameter {name} missing from {path}")
+        view.copy_(model_sd[name].detach().to(torch.float32).reshape(view.shape))
+    return flat

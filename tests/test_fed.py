@@ -474,3 +474,44 @@ def test_watchdog_survives_killed_rank(tiny_cfg, tmp_path):
     assert out["failures_r1"] == 1.0, "dead rank's client counts as a failure"
     assert out["failures_r2"] == 0.0
     assert torch.isfinite(out["params"]).all()
+
+
+def test_restore_cent_from_composer_pt(tiny_cfg, tmp_path):
+    """VERDICT missing #3: a federated run bootstraps from a Composer-keyed
+    ep{e}-ba{b}-rank0.pt checkpoint, params bit-equal to the source
+    (reference init_utils.py:43-125)."""
+    from photon_amd.models import build_model
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    # write a synthetic Composer checkpoint from a randomly-initialized model
+    torch.manual_seed(99)
+    src_model = build_model(cfg["llm_config"])
+    for p in src_model.parameters():
+        p.data.add_(torch.randn_like(p) * 0.01)
+    cent_dir = tmp_path / "cent_run"
+    cent_dir.mkdir(parents=True)
+    torch.save(
+        {"state": {"model": src_model.state_dict(),
+                   "optimizers": {}, "timestamp": {"batch": 120}}},
+        cent_dir / "ep0-ba120-rank0.pt",
+    )
+    torch.save(
+        {"state": {"model": {k: v * 0 for k, v in
+                             src_model.state_dict().items()}}},
+        cent_dir / "ep0-ba60-rank0.pt",
+    )
+
+    cfg["photon"]["restore_cent_run_uuid"] = "cent_run"
+    cfg["photon"]["restore_cent_run_batches"] = 120
+    cfg["photon"]["checkpoint"] = False
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.initialize()
+    ref = FlatParams(src_model).copy_from_model(src_model)
+    assert torch.equal(srv.strategy.params, ref.flat)
+    # and the desired-batches selector is honored (ba60 is all-zeros)
+    cfg2 = copy.deepcopy(cfg)
+    cfg2["photon"]["restore_cent_run_batches"] = 60
+    srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
+    srv2.initialize()
+    assert float(srv2.strategy.params.abs().sum()) == 0.0
