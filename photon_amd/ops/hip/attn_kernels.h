@@ -48,8 +48,17 @@ constexpr int KB = 32;           // keys per kv tile
 #define ATT_DKDV_MINWAVES_D128 1
 #endif
 
+// XOR swizzle field per row: a nibble HALF-SWAP of row&15, found by
+// exhaustive search over GF(2)-linear tables (scripts/swz_search.py):
+// zero LDS bank conflicts simultaneously for the b128 column reads, the
+// ds_read_b64_tr_b16 transpose reads AND the b128 stores at D=64 and
+// D=128. The identity field ((row&15)<<4) used through r01 left the tr16
+// reads 2-way conflicted at D64 and heavily conflicted at D128 (measured
+// SQ_LDS_BANK_CONFLICT 10-13% of wave cycles).
 DEV_INLINE unsigned swz(unsigned byte, int row) {
-  return byte ^ (((unsigned)row & 15u) << 4);
+  const unsigned r = (unsigned)row & 15u;
+  const unsigned t = ((r & 3u) << 2) | (r >> 2);
+  return byte ^ (t << 4);
 }
 
 DEV_INLINE unsigned cvt_pk_bf16(float lo, float hi) {

@@ -1,0 +1,121 @@
+"""Search 16-entry XOR tables (byte ^= T[row&15]<<4) minimizing LDS bank
+conflicts across the three access patterns of attn_kernels.h."""
+import itertools
+
+BANKS = 64
+
+def bank_dwords(byte, ndw):
+    d0 = (byte >> 2) & 63
+    return [(d0 + i) % BANKS for i in range(ndw)]
+
+def group_cost(addr_ndw):  # [(byte, ndwords)] for lanes in one group
+    used = {}
+    for byte, ndw in addr_ndw:
+        for i in range(ndw):
+            b = ((byte >> 2) + i) % BANKS
+            a = (byte & ~3) + 4 * i
+            used.setdefault(b, set()).add(a)
+    return max((len(s) for s in used.values()), default=1) - 1
+
+B128_READ_GROUPS = [
+    [0,1,2,3,12,13,14,15,20,21,22,23,24,25,26,27],
+    [4,5,6,7,8,9,10,11,16,17,18,19,28,29,30,31],
+]
+B128_READ_GROUPS += [[x+32 for x in g] for g in B128_READ_GROUPS]
+TR16_GROUPS = [list(range(0,32)), list(range(32,64))]
+W128_GROUPS = [list(range(8*i, 8*i+8)) for i in range(8)]
+
+def qk_cost(T, D):
+    RS = 2*D
+    total = 0
+    for sub in range(2):
+        for kk in range(D//16):
+            def addr(L):
+                row = sub*32 + (L & 31)
+                byte = row*RS + kk*32 + (L >> 5)*16
+                return byte ^ (T[row & 15] << 4)
+            for g in B128_READ_GROUPS:
+                total += group_cost([(addr(L), 4) for L in g])
+    return total
+
+def tr_cost(T, D):
+    RS = 2*D
+    total = 0
+    for sub in range(2):
+        for s16 in range(2):
+            for rd in range(2):
+                for db in range(D//32):
+                    def addr(L):
+                        tj, tg1, hi = L & 15, (L>>4)&1, L>>5
+                        key0 = 8*hi + 4*rd + (tj >> 2)
+                        row = key0 + 16*s16 + 32*sub
+                        dhc0 = db*32 + 16*tg1 + 4*(tj & 3)
+                        byte = row*RS + dhc0*2
+                        return byte ^ (T[row & 15] << 4)
+                    for g in TR16_GROUPS:
+                        total += group_cost([(addr(L), 2) for L in g])
+    return total
+
+def write_cost(T, D):
+    RS = 2*D
+    total = 0
+    nchunk = 64*D//8//256
+    for i in range(nchunk):
+        for wave in range(4):
+            def addr(lane):
+                t = wave*64 + lane
+                c = i*256 + t
+                row = c // (D//8)
+                col = (c % (D//8))*8
+                byte = row*RS + col*2
+                return byte ^ (T[row & 15] << 4)
+            for g in W128_GROUPS:
+                total += group_cost([(addr(L), 4) for L in g])
+    return total
+
+def cost(T):
+    s = 0
+    for D in (64, 128):
+        s += qk_cost(T, D) + 2*tr_cost(T, D) + write_cost(T, D)
+    return s
+
+ident = [r for r in range(16)]
+print("identity cost:", cost(ident),
+      "| qk64", qk_cost(ident,64), "tr64", tr_cost(ident,64), "w64", write_cost(ident,64),
+      "| qk128", qk_cost(ident,128), "tr128", tr_cost(ident,128), "w128", write_cost(ident,128))
+
+# exhaustive over GF(2) linear maps T[r] = bits of M @ r (4x4 binary)
+best = (cost(ident), tuple(ident), "identity")
+import random
+def linmap(m):
+    # m: 4 ints (columns) -> T[r] = xor of columns where bit set
+    T = []
+    for r in range(16):
+        v = 0
+        for b in range(4):
+            if r >> b & 1:
+                v ^= m[b]
+        T.append(v)
+    return T
+for m in itertools.product(range(16), repeat=4):
+    T = linmap(m)
+    c = cost(T)
+    if c < best[0]:
+        best = (c, tuple(T), f"linear{m}")
+print("best linear:", best[0], best[2], best[1])
+T = list(best[1])
+for D in (64,128):
+    print(f"D{D}: qk {qk_cost(T,D)} tr {tr_cost(T,D)} w {write_cost(T,D)}")
+# random affine/permutation polish
+rng = random.Random(7)
+curT = list(best[1]); curc = best[0]
+for it in range(20000):
+    T2 = list(curT)
+    i = rng.randrange(16)
+    T2[i] = rng.randrange(16)
+    c2 = cost(T2)
+    if c2 <= curc:
+        curT, curc = T2, c2
+print("polished:", curc, tuple(curT))
+for D in (64,128):
+    print(f"D{D}: qk {qk_cost(curT,D)} tr {tr_cost(curT,D)} w {write_cost(curT,D)}")
