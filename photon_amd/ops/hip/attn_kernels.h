@@ -127,38 +127,30 @@ DEV_INLINE float exp2_fast(float x) { return __expf(x * LN2); }
 DEV_INLINE float exp2_fast(float x) { return __builtin_amdgcn_exp2f(x); }
 #endif
 
-// Staging address scheme: thread-fixed 32-bit chunk offsets (voff) + a
-// per-tile uniform 32-bit offset (soff) added to a 64-bit base — ONE
-// strength-reduced add per load instead of the old per-chunk 64-bit
-// row*stride chains. (A raw-buffer-descriptor version with hardware OOB
-// zeroing returned corrupted tiles on gfx950/ROCm 7.2 and was dropped —
-// see round-2 notes; the plain-pointer form measured equal.)
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-#ifdef ATT_STAGE_RSRC2
-// Buffer-descriptor staging, descriptor REBUILT per tile (soffset folded
-// into the base; the SGPR-soffset form returned corrupted tiles on this
-// stack). num_records clamps rows >= S to hardware zeros.
-DEV_INLINE bf16x8 stage_load16x(const __bf16* base, long extent, int voffset,
-                                int soffset) {
-  auto rsrc = __builtin_amdgcn_make_buffer_rsrc(
-      (void*)((const char*)base + (unsigned)soffset), 0,
-      (int)(extent - soffset), 0);
+// Buffer-descriptor (SRSRC, guide T8) staging: scalar 128-bit descriptor +
+// thread-fixed 32-bit voffset + per-tile SGPR soffset — zero per-chunk
+// address VALU, and num_records clamps rows >= S to hardware ZEROS (no
+// bounds compares, no tail branches).
+//
+// flags MUST carry a valid DATA_FORMAT (0x27FAC = dst_sel XYZW | NFMT
+// float | DFMT 32): with flags=0 the FORMAT field is invalid and EVERY
+// load returns zeros — measured, scripts/rsrc_micro.hip (this silently
+// corrupted a whole round of kernels while LOOKING 15% faster thanks to
+// the zero-data DVFS bonus).
+DEV_INLINE __amdgpu_buffer_rsrc_t make_rsrc(const __bf16* base,
+                                            long num_bytes) {
+  return __builtin_amdgcn_make_buffer_rsrc((void*)base, (short)0,
+                                           (int)num_bytes, 0x27FAC);
+}
+
+DEV_INLINE bf16x8 rsrc_load16(__amdgpu_buffer_rsrc_t rsrc, int voffset,
+                              int soffset) {
   union { f32x4 f; bf16x8 b; } u;
-  u.f = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voffset, 0, 0);
+  u.f = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voffset, soffset, 0);
   return u.b;
 }
-#else
-DEV_INLINE bf16x8 stage_load16x(const __bf16* base, long extent, int voffset,
-                                int soffset) {
-  // single 32-bit offset added to the uniform base: lowers to the
-  // global_load saddr form (scalar 64-bit base + 32-bit VGPR offset),
-  // keeping per-chunk addresses out of the VGPR file
-  (void)extent;
-  const unsigned off = (unsigned)voffset + (unsigned)soffset;
-  return *(const bf16x8*)((const char*)base + off);
-}
-#endif
 
 DEV_INLINE float log2_fast(float x) { return __builtin_amdgcn_logf(x); }
 
@@ -312,9 +304,9 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   // (no per-chunk bounds, no 64-bit address chain).
   constexpr int NCHUNK = KBF * D / 8 / ATT_BLOCK;
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
-  const __bf16* kbase = k + ibase;
-  const __bf16* vbase = v + ibase;
   const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
+  const auto krs = make_rsrc(k + ibase, ext_kv);
+  const auto vrs = make_rsrc(v + ibase, ext_kv);
   int voff[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -326,24 +318,11 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   auto stage_load = [&](int t) {
     const int so = t * tile_soff;
 #ifndef ABENCH_NO_LOAD
-    if ((t + 1) * KBF <= S) {  // uniform: full tile, no bounds anywhere
 #pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
-        v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
-      }
-    } else {  // tail tile: zero-fill rows >= S (never read past the tensor)
-#pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
-        if (t * KBF + row < S) {
-          k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
-          v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
-        } else {
-          k_stage[i] = bf16x8{};
-          v_stage[i] = bf16x8{};
-        }
-      }
+    for (int i = 0; i < NCHUNK; ++i) {
+      // rows >= S read hardware zeros via num_records — no tail branch
+      k_stage[i] = rsrc_load16(krs, voff[i], so);
+      v_stage[i] = rsrc_load16(vrs, voff[i], so);
     }
 #else
 #pragma unroll
@@ -659,9 +638,9 @@ void attn_bwd_dq_kernel(
   // SRSRC staging (cf. fwd): hardware-zero OOB rows, no bounds compares
   constexpr int NCHUNK = KBQ * D / 8 / ATT_BLOCK;
   bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
-  const __bf16* kbase = k + ibase;
-  const __bf16* vbase = v + ibase;
   const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
+  const auto krs = make_rsrc(k + ibase, ext_kv);
+  const auto vrs = make_rsrc(v + ibase, ext_kv);
   int voff[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -671,24 +650,10 @@ void attn_bwd_dq_kernel(
   const int tile_soff = (int)(KBQ * rs_i * 2);
   auto stage_load = [&](int t) {
     const int so = t * tile_soff;
-    if ((t + 1) * KBQ <= S) {  // uniform fast path (cf. fwd)
 #pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
-        v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
-        if (t * KBQ + row < S) {
-          k_stage[i] = stage_load16x(kbase, ext_kv, voff[i], so);
-          v_stage[i] = stage_load16x(vbase, ext_kv, voff[i], so);
-        } else {
-          k_stage[i] = bf16x8{};
-          v_stage[i] = bf16x8{};
-        }
-      }
+    for (int i = 0; i < NCHUNK; ++i) {
+      k_stage[i] = rsrc_load16(krs, voff[i], so);
+      v_stage[i] = rsrc_load16(vrs, voff[i], so);
     }
   };
   auto stage_write = [&](int b) {
@@ -917,10 +882,8 @@ void attn_bwd_dkdv_kernel(
 
   // SRSRC staging (cf. fwd). q and dout have separate strides (packed QKV
   // input vs contiguous dout) -> separate descriptors/voffsets.
-  const __bf16* qbase = q + ibase;
-  const __bf16* obase_p = dout + obase;
-  const long ext_q = ((long)(S - 1) * rs_i + D) * 2;
-  const long ext_o = ((long)(S - 1) * rs_o + D) * 2;
+  const auto qrs = make_rsrc(q + ibase, ((long)(S - 1) * rs_i + D) * 2);
+  const auto ors = make_rsrc(dout + obase, ((long)(S - 1) * rs_o + D) * 2);
   int voff_q[NCHUNK], voff_o[NCHUNK];
 #pragma unroll
   for (int i = 0; i < NCHUNK; ++i) {
@@ -934,24 +897,10 @@ void attn_bwd_dkdv_kernel(
   const int tsoff_o = (int)(QTF * rs_o * 2);
 
   auto stage_load = [&](int t) {
-    if ((t + 1) * QTF <= S) {  // uniform fast path (cf. fwd)
 #pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        q_stage[i] = stage_load16x(qbase, ext_q, voff_q[i], t * tsoff_q);
-        do_stage[i] = stage_load16x(obase_p, ext_o, voff_o[i], t * tsoff_o);
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < NCHUNK; ++i) {
-        const int row = (i * ATT_BLOCK + (int)threadIdx.x) / (D / 8);
-        if (t * QTF + row < S) {
-          q_stage[i] = stage_load16x(qbase, ext_q, voff_q[i], t * tsoff_q);
-          do_stage[i] = stage_load16x(obase_p, ext_o, voff_o[i], t * tsoff_o);
-        } else {
-          q_stage[i] = bf16x8{};
-          do_stage[i] = bf16x8{};
-        }
-      }
+    for (int i = 0; i < NCHUNK; ++i) {
+      q_stage[i] = rsrc_load16(qrs, voff_q[i], t * tsoff_q);
+      do_stage[i] = rsrc_load16(ors, voff_o[i], t * tsoff_o);
     }
     if (threadIdx.x < QTF) {
       const long qi = (long)t * QTF + threadIdx.x;

@@ -1,4 +1,7 @@
 #include <hip/hip_runtime.h>
+#ifndef RSRC_FLAGS
+#define RSRC_FLAGS 0
+#endif
 #include <cstdio>
 #include <vector>
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -6,7 +9,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 __global__ void k_rsrc(const char* src, char* dst, int extent, int tiles, int tile_bytes) {
   for (int t = 0; t < tiles; ++t) {
     const int so = t * tile_bytes;
-    auto rsrc = __builtin_amdgcn_make_buffer_rsrc((void*)(src + so), (short)0, extent - so, 0);
+    auto rsrc = __builtin_amdgcn_make_buffer_rsrc((void*)(src + so), (short)0, extent - so, RSRC_FLAGS);
     for (int i = 0; i < 2; ++i) {
       int voff = i * 4096 + threadIdx.x * 16;
       f32x4 v = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voff, 0, 0);
@@ -15,7 +18,7 @@ __global__ void k_rsrc(const char* src, char* dst, int extent, int tiles, int ti
   }
 }
 __global__ void k_rsrc_soff(const char* src, char* dst, int extent, int tiles, int tile_bytes) {
-  auto rsrc = __builtin_amdgcn_make_buffer_rsrc((void*)src, (short)0, extent, 0);
+  auto rsrc = __builtin_amdgcn_make_buffer_rsrc((void*)src, (short)0, extent, RSRC_FLAGS);
   for (int t = 0; t < tiles; ++t) {
     const int so = t * tile_bytes;
     for (int i = 0; i < 2; ++i) {
