@@ -129,24 +129,38 @@ class Trainer:
         n = len(batches)
         total_loss_t = None
         if self.use_hip_graphs and self.device.type == "cuda":
-            # graph path: grads are captured tensors — zero in place
-            if self._graphed is None:
-                from .graphs import GraphedMicrobatch
+            # graph path: grads are captured tensors — zero in place.
+            # Capture failure (unsupported op under capture, allocator
+            # state) falls back to eager PERMANENTLY and loudly — never
+            # abort a run over an optimization.
+            try:
+                if self._graphed is None:
+                    from .graphs import GraphedMicrobatch
 
-                ids0 = batches[0]["input_ids"]
-                self._graphed = GraphedMicrobatch(
-                    self.model, self.autocast, float(self.grad_accum),
-                    ids0.shape[0], ids0.shape[1], self.device,
-                )
-            else:
-                self._graphed._zero_grads()
-            for mb in batches:
-                ids = mb["input_ids"].to(self.device, non_blocking=True)
-                loss = self._graphed.run(ids)
-                total_loss_t = (
-                    loss.clone() if total_loss_t is None else total_loss_t + loss
-                )
-        else:
+                    ids0 = batches[0]["input_ids"]
+                    self._graphed = GraphedMicrobatch(
+                        self.model, self.autocast, float(self.grad_accum),
+                        ids0.shape[0], ids0.shape[1], self.device,
+                    )
+                else:
+                    self._graphed._zero_grads()
+                for mb in batches:
+                    ids = mb["input_ids"].to(self.device, non_blocking=True)
+                    loss = self._graphed.run(ids)
+                    total_loss_t = (
+                        loss.clone() if total_loss_t is None
+                        else total_loss_t + loss
+                    )
+            except RuntimeError as e:
+                print(f"[graphs] capture/replay failed ({e}); "
+                      "falling back to eager", flush=True)
+                self.use_hip_graphs = False
+                self._graphed = None
+                self.optimizer.zero_grad(set_to_none=True)
+                total_loss_t = None
+        if total_loss_t is None and not (
+            self.use_hip_graphs and self.device.type == "cuda"
+        ):
             self.optimizer.zero_grad(set_to_none=True)
             # ONE autocast region for the whole accumulation loop: the
             # autocast weight-cast cache then converts each fp32 weight to
