@@ -40,10 +40,11 @@ def main() -> None:
     ap.add_argument("--microbatch", type=int, default=32)
     ap.add_argument("--attn", default="flash", choices=["flash", "torch"])
     ap.add_argument("--hip-graphs", action=argparse.BooleanOptionalAction,
-                    default=True,
+                    default=False,
                     help="capture the microbatch fwd+bwd in a hipGraph "
                          "(falls back to eager on capture failure; "
-                         "graphs+masters measured +2%% vs eager+autocast)")
+                         "box-dependent +-1-2%% at 125M, -2%% at 1B -> "
+                         "off by default)")
     ap.add_argument("--master-weights", action=argparse.BooleanOptionalAction,
                     default=True,
                     help="bf16 model weights + fp32 optimizer masters "
