@@ -131,10 +131,20 @@ class MPTBlock(nn.Module):
         self.norm_2 = FusedLayerNorm(cfg.d_model, bias=bias)
         self.ffn = MPTMLP(cfg)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = x + self.attn(self.norm_1(x))
-        x = x + self.ffn(self.norm_2(x))
-        return x
+    def forward(self, x: torch.Tensor, pending: torch.Tensor | None = None):
+        """Residual-add/LN fusion flow: ``pending`` is the PREVIOUS
+        block's un-added FFN output (or None for block 0); the add rides
+        the norm_1 kernel. Returns (residual_stream, ffn_out) — the caller
+        (MPTModel) threads ffn_out into the next block / the final norm.
+        Eager semantics are identical: x+pending -> norm_1 -> attn -> add
+        -> norm_2 -> ffn."""
+        if pending is not None:
+            x, n1 = self.norm_1.forward_add(x, pending)
+        else:
+            n1 = self.norm_1(x)
+        a = self.attn(n1)
+        x, n2 = self.norm_2.forward_add(x, a)
+        return x, self.ffn(n2)
 
 
 class MPTModel(nn.Module):
@@ -157,9 +167,12 @@ class MPTModel(nn.Module):
         # LN/CE kernels still accumulate in fp32.
         if torch.is_autocast_enabled(x.device.type):
             x = x.to(torch.get_autocast_dtype(x.device.type))
+        pending = None
         for block in self.blocks:
-            x = block(x)
-        return self.norm_f(x)
+            x, pending = block(x, pending)
+        # final residual add fused into norm_f
+        _, y = self.norm_f.forward_add(x, pending)
+        return y
 
 
 class MPTCausalLM(nn.Module):

@@ -10,10 +10,12 @@ namespace photon_hip {
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
                                          c10::optional<torch::Tensor> b,
-                                         double eps);
+                                         double eps,
+                                         c10::optional<torch::Tensor> residual);
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
-                                         torch::Tensor rstd);
+                                         torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> dresid);
 torch::Tensor ce_fwd_bwd_inplace(torch::Tensor logits, torch::Tensor targets);
 void adamw_step(std::vector<torch::Tensor> ps, std::vector<torch::Tensor> gs,
                 std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs,
@@ -59,8 +61,14 @@ torch::Tensor fused_linear(torch::Tensor x, torch::Tensor w,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   using namespace photon_hip;
-  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm forward");
-  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm backward");
+  m.def("layernorm_fwd", &layernorm_fwd,
+        "fused LayerNorm forward (optional fused residual add)",
+        py::arg("x"), py::arg("w"), py::arg("b"), py::arg("eps"),
+        py::arg("residual") = c10::nullopt);
+  m.def("layernorm_bwd", &layernorm_bwd,
+        "fused LayerNorm backward (optional residual-grad add-through)",
+        py::arg("dy"), py::arg("x"), py::arg("w"), py::arg("mean"),
+        py::arg("rstd"), py::arg("dresid") = c10::nullopt);
   m.def("ce_fwd_bwd_inplace", &ce_fwd_bwd_inplace,
         "fused CE loss + in-place softmax gradient");
   m.def("adamw_step", &adamw_step, "fused multi-tensor AdamW step");

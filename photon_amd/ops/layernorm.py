@@ -33,6 +33,37 @@ class _LayerNormHIP(torch.autograd.Function):
         return dx, dw, (db if ctx.has_bias else None), None
 
 
+class _LayerNormAddHIP(torch.autograd.Function):
+    """Fused residual add + LayerNorm: forward returns (s, y) with
+    s = round_bf16(x + r) written by the LN kernel itself (no standalone
+    add kernel) and y = LN(s); backward folds the incoming residual-stream
+    gradient ds into the LN dx epilogue, so dL/dx = dL/dr needs no extra
+    elementwise add either."""
+
+    @staticmethod
+    def forward(ctx, x, r, weight, bias, eps):
+        # norm_f discards s: receive None instead of a materialized zeros
+        # grad so the kernel skips the add-through entirely
+        ctx.set_materialize_grads(False)
+        ext = hip_ext()
+        y, mean, rstd, s = ext.layernorm_fwd(x, weight, bias, eps, r)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        ctx.has_bias = bias is not None
+        return s, y
+
+    @staticmethod
+    def backward(ctx, ds, dy):
+        s, weight, mean, rstd = ctx.saved_tensors
+        ext = hip_ext()
+        if dy is None:  # y unused (cannot happen in the model, but be safe)
+            dy = torch.zeros_like(s)
+        dx, dw, db = ext.layernorm_bwd(
+            dy.contiguous(), s, weight, mean, rstd,
+            ds.contiguous() if ds is not None else None,
+        )
+        return dx, dx, dw, (db if ctx.has_bias else None), None
+
+
 class FusedLayerNorm(nn.Module):
     def __init__(self, normalized_shape: int, eps: float = 1e-5, bias: bool = True):
         super().__init__()
@@ -40,6 +71,19 @@ class FusedLayerNorm(nn.Module):
         self.eps = eps
         self.weight = nn.Parameter(torch.ones(normalized_shape))
         self.bias = nn.Parameter(torch.zeros(normalized_shape)) if bias else None
+
+    def forward_add(self, x: torch.Tensor, r: torch.Tensor):
+        """(s, y) = (x + r, LN(x + r)) with the add fused into the LN
+        kernel on GPU; eager fallback elsewhere."""
+        d = self.normalized_shape[0]
+        if (d % 256 == 0 and d <= 4096 and use_hip(x)
+                and x.dtype == r.dtype and x.dtype != torch.float32):
+            return _LayerNormAddHIP.apply(
+                x.contiguous(), r.contiguous(), self.weight, self.bias,
+                self.eps,
+            )
+        s = x + r
+        return s, self.forward(s)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         d = self.normalized_shape[0]

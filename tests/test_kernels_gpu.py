@@ -514,3 +514,70 @@ def test_cpp_fused_linear_grads(dev, ext):
         assert relw < 0.03, (autocast_mode, relw)
         relb = (b.grad - br.grad).abs().max() / br.grad.abs().max()
         assert relb < 0.03, (autocast_mode, relb)
+
+
+def test_layernorm_fused_residual_add(dev, ext):
+    """Fused s=x+r + LN forward and the residual-grad add-through backward
+    vs the eager two-op reference (bf16, bit-matching rounding of s)."""
+    torch.manual_seed(51)
+    N, D = 512, 768
+    x = (torch.randn(N, D, device=dev) * 0.5).to(torch.bfloat16).requires_grad_()
+    r = (torch.randn(N, D, device=dev) * 0.5).to(torch.bfloat16).requires_grad_()
+    w = torch.randn(D, device=dev).requires_grad_()
+    b = torch.randn(D, device=dev).requires_grad_()
+    from photon_amd.ops.layernorm import _LayerNormAddHIP
+
+    s, y = _LayerNormAddHIP.apply(x, r, w, b, 1e-5)
+    ds = (torch.randn_like(s.float()) * 0.1).to(torch.bfloat16)
+    dy = (torch.randn_like(y.float()) * 0.1).to(torch.bfloat16)
+    torch.autograd.backward([s, y], [ds, dy])
+
+    xr = x.detach().clone().requires_grad_()
+    rr = r.detach().clone().requires_grad_()
+    wr = w.detach().clone().requires_grad_()
+    br = b.detach().clone().requires_grad_()
+    sr = xr + rr
+    yr = torch.nn.functional.layer_norm(sr.float(), (D,), wr, br, 1e-5)
+    torch.autograd.backward([sr, yr], [ds, dy.float()])
+
+    assert torch.equal(s, (x.detach() + r.detach())), "s must be the bf16 sum"
+    assert (y.float() - yr).abs().max() < 3e-2
+    assert (x.grad.float() - xr.grad.float()).abs().max() < 3e-2
+    assert torch.equal(x.grad, r.grad), "residual pair shares the gradient"
+    assert (w.grad - wr.grad).abs().max() / wr.grad.abs().max() < 0.03
+    assert (b.grad - br.grad).abs().max() / br.grad.abs().max() < 0.03
+
+
+def test_model_block_fused_flow_matches_eager(dev, ext):
+    """The pending-add block flow must produce the same loss/grads as the
+    naive x + attn(ln(x)) composition (GPU bf16)."""
+    from photon_amd.models import build_model
+
+    cfg = {
+        "model": {"d_model": 256, "n_heads": 4, "n_layers": 3,
+                  "expansion_ratio": 2, "max_seq_len": 128,
+                  "vocab_size": 512,
+                  "attn_config": {"attn_impl": "flash"}}
+    }
+    torch.manual_seed(52)
+    m = build_model(cfg).to(dev).to(torch.bfloat16)
+    ids = torch.randint(0, 512, (2, 128), device=dev)
+    out = m(ids, labels=ids)
+    out["loss"].backward()
+    # eager reference: run the same weights through the naive composition
+    import copy as _copy
+
+    m2 = build_model(cfg).to(dev).to(torch.bfloat16)
+    m2.load_state_dict(m.state_dict())
+    x = m2.transformer.wte(ids)
+    for blk in m2.transformer.blocks:
+        x = x + blk.attn(blk.norm_1(x))
+        x = x + blk.ffn(blk.norm_2(x))
+    x = m2.transformer.norm_f(x)
+    logits = torch.nn.functional.linear(x, m2.transformer.wte.weight)
+    ref_loss = torch.nn.functional.cross_entropy(
+        logits[:, :-1].reshape(-1, 512).float(), ids[:, 1:].reshape(-1)
+    )
+    assert abs(float(out["loss"]) - float(ref_loss)) < 5e-2
+    for p in m.parameters():
+        assert p.grad is None or torch.isfinite(p.grad).all()
