@@ -59,13 +59,15 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x,
         const int i = c * COLS_PER_WAVE + lane * 4;
         floatx4 a = load4<T>(xr + i);
         floatx4 bv = load4<T>(rr + i);
-        // round to T first: stats must see the stored values
+        // round through T first (store4 does the T-rounding), then re-read
+        // the rounded values: stats must see exactly the stored stream
         floatx4 o;
-        o.x = (float)(T)(a.x + bv.x);
-        o.y = (float)(T)(a.y + bv.y);
-        o.z = (float)(T)(a.z + bv.z);
-        o.w = (float)(T)(a.w + bv.w);
+        o.x = a.x + bv.x;
+        o.y = a.y + bv.y;
+        o.z = a.z + bv.z;
+        o.w = a.w + bv.w;
         store4<T>(so + i, o);
+        o = load4<T>(so + i);
         s += o.x + o.y + o.z + o.w;
       }
       xr = so;  // variance + normalize passes read the sum
