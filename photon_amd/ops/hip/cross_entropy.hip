@@ -6,12 +6,14 @@
 // The surrounding GEMMs (hipBLASLt) consume that in-place gradient, so the
 // full-precision logits never round-trip to HBM twice.
 //
-// v2 (profiles/r01: v1 ran at 1.4 TB/s): branch-free three-pass structure —
-// (1) pure vector max, (2) e = exp(v - max) summed AND stored in place
-// (ONE exp per element instead of v1's two plus a divergent online-max
-// branch), (3) scale by 1/sum. A row is 100 KB bf16: pass 2/3 re-reads hit
-// the XCD L2, so HBM sees one read + one write per element. All accesses
-// are 8-element vectors (v1 stored scalar 2-byte elements).
+// v3: TWO passes. v2's three-pass structure assumed the 100 KB row stayed
+// in the XCD L2 between passes, but at full occupancy 256 resident blocks
+// per XCD hold 25 MB of rows against a 4 MB L2 — every pass went to HBM
+// (measured: 4.5 ms per [65536, 50368] call = the 3R+2W HBM bound).
+// Pass A reads the row ONCE with an online max+sum (rescale-by-exp on max
+// growth, then a two-step block combine: max first, then sums scaled by
+// exp(m_i - gm)); pass B re-reads and writes the gradient in one sweep.
+// 2R + 1W = the minimum for an in-place grad without holding the row.
 
 #include "host_common.h"
 
@@ -27,55 +29,54 @@ __global__ __launch_bounds__(BLOCK) void ce_fwd_bwd_kernel(
   const long tgt = targets[row];
   const float logit_tgt = load_f32<T>(lr, tgt);
 
-  // pass 1: max (vector loads, no exp, no branches)
-  float m = -1e30f;
+  // pass A: ONE read, online max + rescaled sum per thread
+  float m = -1e30f, s = 0.f;
+#pragma unroll 2
   for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
     floatx4 a = load4<T>(lr + i);
     floatx4 b = load4<T>(lr + i + 4);
-    m = fmaxf(m, fmaxf(fmaxf(a.x, a.y), fmaxf(a.z, a.w)));
-    m = fmaxf(m, fmaxf(fmaxf(b.x, b.y), fmaxf(b.z, b.w)));
-  }
-  for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK)
-    m = fmaxf(m, load_f32<T>(lr, i));
-  const float gm = block_reduce_max(m, scratch);
-
-  // pass 2: e = exp(v - gm), accumulate sum, store e in place (L2-resident)
-  float s = 0.f;
-  for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
-    floatx4 a = load4<T>(lr + i);
-    floatx4 b = load4<T>(lr + i + 4);
-    floatx4 ea, eb;
-    ea.x = __expf(a.x - gm); ea.y = __expf(a.y - gm);
-    ea.z = __expf(a.z - gm); ea.w = __expf(a.w - gm);
-    eb.x = __expf(b.x - gm); eb.y = __expf(b.y - gm);
-    eb.z = __expf(b.z - gm); eb.w = __expf(b.w - gm);
-    s += ea.x + ea.y + ea.z + ea.w + eb.x + eb.y + eb.z + eb.w;
-    store4<T>(lr + i, ea);
-    store4<T>(lr + i + 4, eb);
+    const float lm = fmaxf(
+        fmaxf(fmaxf(a.x, a.y), fmaxf(a.z, a.w)),
+        fmaxf(fmaxf(b.x, b.y), fmaxf(b.z, b.w)));
+    if (lm > m) {
+      s *= __expf(m - lm);  // exp(-huge) == 0 on the first tile
+      m = lm;
+    }
+    s += __expf(a.x - m) + __expf(a.y - m) + __expf(a.z - m) +
+         __expf(a.w - m) + __expf(b.x - m) + __expf(b.y - m) +
+         __expf(b.z - m) + __expf(b.w - m);
   }
   for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK) {
-    float e = __expf(load_f32<T>(lr, i) - gm);
-    s += e;
-    store_f32<T>(lr, i, e);
+    const float v = load_f32<T>(lr, i);
+    if (v > m) {
+      s *= __expf(m - v);
+      m = v;
+    }
+    s += __expf(v - m);
   }
-  const float gs = block_reduce_sum(s, scratch);
+  // combine (m, s) pairs: global max first, then sums scaled into it
+  const float gm = block_reduce_max(m, scratch);
+  __syncthreads();  // scratch reuse between the two reductions
+  const float gs = block_reduce_sum(s * __expf(m - gm), scratch);
   const float inv = 1.f / gs;
   if (threadIdx.x == 0) {
     losses[row] = gm + __logf(gs) - logit_tgt;
   }
-  __syncthreads();
 
-  // pass 3: p = e * inv (- onehot); vector stores
+  // pass B: re-read, write p = exp(v - gm)/gs (- onehot) in one sweep
+#pragma unroll 2
   for (long i = (long)threadIdx.x * 8; i + 7 < V; i += (long)BLOCK * 8) {
     floatx4 a = load4<T>(lr + i);
     floatx4 b = load4<T>(lr + i + 4);
-    a.x *= inv; a.y *= inv; a.z *= inv; a.w *= inv;
-    b.x *= inv; b.y *= inv; b.z *= inv; b.w *= inv;
+    a.x = __expf(a.x - gm) * inv; a.y = __expf(a.y - gm) * inv;
+    a.z = __expf(a.z - gm) * inv; a.w = __expf(a.w - gm) * inv;
+    b.x = __expf(b.x - gm) * inv; b.y = __expf(b.y - gm) * inv;
+    b.z = __expf(b.z - gm) * inv; b.w = __expf(b.w - gm) * inv;
     store4<T>(lr + i, a);
     store4<T>(lr + i + 4, b);
   }
   for (long i = (V / 8) * 8 + threadIdx.x; i < V; i += BLOCK) {
-    store_f32<T>(lr, i, load_f32<T>(lr, i) * inv);
+    store_f32<T>(lr, i, __expf(load_f32<T>(lr, i) - gm) * inv);
   }
   __syncthreads();
   if (threadIdx.x == 0) {
