@@ -48,17 +48,29 @@ constexpr int KB = 32;           // keys per kv tile
 #define ATT_DKDV_MINWAVES_D128 1
 #endif
 
-// XOR swizzle field per row: a nibble HALF-SWAP of row&15, found by
-// exhaustive search over GF(2)-linear tables (scripts/swz_search.py):
-// zero LDS bank conflicts simultaneously for the b128 column reads, the
-// ds_read_b64_tr_b16 transpose reads AND the b128 stores at D=64 and
-// D=128. The identity field ((row&15)<<4) used through r01 left the tr16
-// reads 2-way conflicted at D64 and heavily conflicted at D128 (measured
-// SQ_LDS_BANK_CONFLICT 10-13% of wave cycles).
-DEV_INLINE unsigned swz(unsigned byte, int row) {
-  const unsigned r = (unsigned)row & 15u;
-  const unsigned t = ((r & 3u) << 2) | (r >> 2);
-  return byte ^ (t << 4);
+// XOR swizzle field per row, found by exhaustive search over GF(2)-linear
+// tables (scripts/swz_search.py): zero LDS bank conflicts simultaneously
+// for the b128 column reads, the ds_read_b64_tr_b16 transpose reads AND
+// the b128 stores, at both D=64 and D=128 (the r01 identity field left
+// the tr16 reads 2-way conflicted at D64, worse at D128 — measured
+// SQ_LDS_BANK_CONFLICT 10-13% of wave cycles, now 2-3%).
+//
+// ROW-PRESERVING constraint: the field must not flip byte bits >= the row
+// stride so the swizzle is an XOR involution WITHIN each row — that makes
+// the LDS-DMA source permutation trivially invertible. At D=64 (128 B
+// rows) that limits the field to 3 bits; the 4-bit half-swap is fine at
+// D=128 (256 B rows).
+DEV_INLINE unsigned swz_field(unsigned r, bool d64) {
+  if (d64) {
+    // best zero-conflict 3-bit GF(2)-linear map: t = {bit1->4, bit2->1,
+    // bit3->2} of row&15
+    return (((r >> 1) & 1u) << 2) | ((r >> 2) & 1u) | (((r >> 3) & 1u) << 1);
+  }
+  return ((r & 3u) << 2) | (r >> 2);  // nibble half-swap
+}
+
+DEV_INLINE unsigned swz(unsigned byte, int row, bool d64) {
+  return byte ^ (swz_field((unsigned)row & 15u, d64) << 4);
 }
 
 DEV_INLINE unsigned cvt_pk_bf16(float lo, float hi) {
@@ -93,19 +105,22 @@ DEV_INLINE bf16x8 pack_bfrag(const float* p, int rb) {
 // swizzle. byte = row*rowstride + coloff must be 16B-aligned pre-swizzle.
 DEV_INLINE bf16x8 lds_frag(const __bf16* img, int row, int rowstride_b,
                            int coloff_b) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
+                      rowstride_b == 128);
   return *(const bf16x8*)((const char*)img + byte);
 }
 
 DEV_INLINE void lds_store16(__bf16* img, int row, int rowstride_b,
                             int coloff_b, bf16x8 v) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
+                      rowstride_b == 128);
   *(bf16x8*)((char*)img + byte) = v;
 }
 
 DEV_INLINE void lds_store2(__bf16* img, int row, int rowstride_b,
                            int coloff_b, __bf16 v) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row);
+  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
+                      rowstride_b == 128);
   *(__bf16*)((char*)img + byte) = v;
 }
 
@@ -150,6 +165,21 @@ DEV_INLINE bf16x8 rsrc_load16(__amdgpu_buffer_rsrc_t rsrc, int voffset,
   union { f32x4 f; bf16x8 b; } u;
   u.f = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voffset, soffset, 0);
   return u.b;
+}
+
+// LDS-DMA: one wave-instruction moves 1 KiB (16 B/lane) global -> LDS with
+// NO data registers and no ds_write issue cost (buffer_load_dwordx4 ..lds).
+// The LDS destination is linear in lanes, so the ROW-PRESERVING XOR
+// swizzle is applied to the SOURCE offsets instead (involution per row).
+DEV_INLINE void lds_dma16(__amdgpu_buffer_rsrc_t rsrc, char* lds_dest,
+                          int voffset, int soffset) {
+  typedef __attribute__((address_space(3))) void* lds_vp;
+  __builtin_amdgcn_raw_ptr_buffer_load_lds(
+      rsrc, (lds_vp)lds_dest, 16, voffset, soffset, 0, 0);
+}
+
+DEV_INLINE void vm_wait0() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 }
 
 DEV_INLINE float log2_fast(float x) { return __builtin_amdgcn_logf(x); }
@@ -299,58 +329,39 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     alibi2[r] = slope2 * (float)pat;
   }
 
-  // SRSRC register staging (T8): thread-fixed 32-bit voffset + per-tile
-  // scalar soffset; num_records sized so rows >= S read hardware zeros
-  // (no per-chunk bounds, no 64-bit address chain).
-  constexpr int NCHUNK = KBF * D / 8 / ATT_BLOCK;
-  bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
+  // LDS-DMA staging: NW 1-KiB windows per wave per tensor go straight
+  // from HBM to LDS (no staging registers, no ds_write issue cost, rows
+  // >= S hardware-zeroed by num_records). Source offsets carry the
+  // swizzle (involution within each row).
+  constexpr int NW = KBF * D * 2 / 1024 / WAVES;
   const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
   const auto krs = make_rsrc(k + ibase, ext_kv);
   const auto vrs = make_rsrc(v + ibase, ext_kv);
-  int voff[NCHUNK];
+  int dma_voff[NW];
 #pragma unroll
-  for (int i = 0; i < NCHUNK; ++i) {
-    const int c = i * ATT_BLOCK + threadIdx.x;
-    voff[i] = (int)(((long)(c / (D / 8)) * rs_i + (c % (D / 8)) * 8) * 2);
+  for (int i = 0; i < NW; ++i) {
+    const int dest = (wave * NW + i) * 1024 + lane * 16;  // image byte
+    const int row = dest / (D * 2);
+    const int src = (int)(dest ^ (swz_field(row & 15u, D == 64) << 4));
+    dma_voff[i] = (int)((long)(src / (D * 2)) * rs_i * 2 + src % (D * 2));
   }
   const int tile_soff = (int)(KBF * rs_i * 2);
 
-  auto stage_load = [&](int t) {
+  auto dma_stage = [&](int t, int b) {
     const int so = t * tile_soff;
 #ifndef ABENCH_NO_LOAD
 #pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      // rows >= S read hardware zeros via num_records — no tail branch
-      k_stage[i] = rsrc_load16(krs, voff[i], so);
-      v_stage[i] = rsrc_load16(vrs, voff[i], so);
+    for (int i = 0; i < NW; ++i) {
+      const int win = (wave * NW + i) * 1024;
+      lds_dma16(krs, (char*)k_img(b) + win, dma_voff[i], so);
+      lds_dma16(vrs, (char*)v_img(b) + win, dma_voff[i], so);
     }
 #else
-#pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      k_stage[i] = bf16x8{};
-      v_stage[i] = bf16x8{};
-    }
     (void)so;
+    (void)b;
 #endif
   };
-  auto stage_write = [&](int b) {
-#pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-#ifndef ABENCH_NO_KSTORE
-      lds_store16(k_img(b), row, D * 2, col * 2, k_stage[i]);
-#endif
-#ifndef ABENCH_NO_VT
-      // V is a ROW image too; PV reads it transposed via lds_tr16.
-      lds_store16(v_img(b), row, D * 2, col * 2, v_stage[i]);
-#endif
-    }
-  };
-
-  stage_load(0);
-  stage_write(0);
+  dma_stage(0, 0);
 
 #if ATT_WAVES == 8 && defined(ATT_SETPRIO)
   // 8-wave workgroups put two waves on each SIMD; the second-dispatched
@@ -371,7 +382,7 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   unsigned qk_addr[D / 16];  // per-kk QK fragment address (sub=0, buf=0)
 #pragma unroll
   for (int kk = 0; kk < D / 16; ++kk) {
-    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq, D == 64);
   }
   unsigned pv_addr[2][D / 32];  // per (rd, db) tr-read address
 #pragma unroll
@@ -380,7 +391,7 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
     for (int db = 0; db < D / 32; ++db) {
       const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
       const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
-      pv_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+      pv_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0, D == 64);
     }
   }
 
@@ -443,10 +454,11 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
   // ---- interior tiles: no masks, no active checks, straight-line ---------
   for (int t = 0; t < t_int; ++t) {
     const int buf = t & 1;
-    if (t + 1 < n_tiles) stage_load(t + 1);  // hide HBM under this tile
+    vm_wait0();  // this tile's DMA landed
 #ifndef ABENCH_NO_BARRIER
-    __syncthreads();  // LDS[buf] writes (prev iter) visible to all waves
+    __syncthreads();  // LDS[buf] visible to all waves; 1-buf reads done
 #endif
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - buf);  // hide HBM under t
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       f32x16 s_acc = qk_mfma(buf, sub);
@@ -484,17 +496,17 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
 #endif  // ABENCH_NO_SOFTMAX
       pv_accum(p, buf, sub);
     }
-    if (t + 1 < n_tiles) stage_write(1 - buf);
   }
 
   // ---- boundary tiles: diagonal/tail masking (<= 2 per block) ------------
   for (int t = t_int; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int kv0 = t * KBF;
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
 #ifndef ABENCH_NO_BARRIER
     __syncthreads();
 #endif
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - buf);
     const bool active = !causal || (kv0 <= my_q_max);
     if (active) {
 #pragma unroll
@@ -530,7 +542,6 @@ __global__ __launch_bounds__(ATT_BLOCK, ATT_FWD_MINWAVES) void attn_fwd_kernel(
         pv_accum(p, buf, sub);
       }
     }
-    if (t + 1 < n_tiles) stage_write(1 - buf);
   }
   __syncthreads();  // protect epilogue smem reuse
 
@@ -635,40 +646,30 @@ void attn_bwd_dq_kernel(
   int t_int = causal ? (q0_block / KBQ) : (S / KBQ);
   if (t_int > n_tiles) t_int = n_tiles;
 
-  // SRSRC staging (cf. fwd): hardware-zero OOB rows, no bounds compares
-  constexpr int NCHUNK = KBQ * D / 8 / ATT_BLOCK;
-  bf16x8 k_stage[NCHUNK], v_stage[NCHUNK];
+  // LDS-DMA staging (cf. fwd): swizzled SOURCE offsets, no staging regs
+  constexpr int NW = KBQ * D * 2 / 1024 / WAVES;
   const long ext_kv = ((long)(S - 1) * rs_i + D) * 2;
   const auto krs = make_rsrc(k + ibase, ext_kv);
   const auto vrs = make_rsrc(v + ibase, ext_kv);
-  int voff[NCHUNK];
+  int dma_voff[NW];
 #pragma unroll
-  for (int i = 0; i < NCHUNK; ++i) {
-    const int c = i * ATT_BLOCK + threadIdx.x;
-    voff[i] = (int)(((long)(c / (D / 8)) * rs_i + (c % (D / 8)) * 8) * 2);
+  for (int i = 0; i < NW; ++i) {
+    const int dest = (wave * NW + i) * 1024 + lane * 16;
+    const int row = dest / (D * 2);
+    const int src = (int)(dest ^ (swz_field(row & 15u, D == 64) << 4));
+    dma_voff[i] = (int)((long)(src / (D * 2)) * rs_i * 2 + src % (D * 2));
   }
   const int tile_soff = (int)(KBQ * rs_i * 2);
-  auto stage_load = [&](int t) {
+  auto dma_stage = [&](int t, int b) {
     const int so = t * tile_soff;
 #pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      k_stage[i] = rsrc_load16(krs, voff[i], so);
-      v_stage[i] = rsrc_load16(vrs, voff[i], so);
+    for (int i = 0; i < NW; ++i) {
+      const int win = (wave * NW + i) * 1024;
+      lds_dma16(krs, (char*)k_img(b) + win, dma_voff[i], so);
+      lds_dma16(vrs, (char*)v_img(b) + win, dma_voff[i], so);
     }
   };
-  auto stage_write = [&](int b) {
-#pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-      lds_store16(k_img(b), row, D * 2, col * 2, k_stage[i]);
-      lds_store16(v_img(b), row, D * 2, col * 2, v_stage[i]);
-    }
-  };
-
-  stage_load(0);
-  stage_write(0);
+  dma_stage(0, 0);
 
   // Hoisted LDS read addresses (cf. fwd): same k/v row-image layout, so
   // one address set serves both images (v_img = k_img + IMG2 immediate).
@@ -677,7 +678,7 @@ void attn_bwd_dq_kernel(
   unsigned qk_addr[D / 16];
 #pragma unroll
   for (int kk = 0; kk < D / 16; ++kk) {
-    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq, D == 64);
   }
   unsigned tr_addr[2][D / 32];
 #pragma unroll
@@ -686,7 +687,7 @@ void attn_bwd_dq_kernel(
     for (int db = 0; db < D / 32; ++db) {
       const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
       const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
-      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0, D == 64);
     }
   }
 
@@ -725,8 +726,9 @@ void attn_bwd_dq_kernel(
   // ---- interior tiles (no masks): p*scale = exp2(sv2 - lse2) ------------
   for (int t = 0; t < t_int; ++t) {
     const int buf = t & 1;
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
     __syncthreads();
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - buf);
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       f32x16 s_acc = f32x16{}, dp_acc = f32x16{};
@@ -745,15 +747,15 @@ void attn_bwd_dq_kernel(
       }
       dsk_mfma(ds, buf, sub);
     }
-    if (t + 1 < n_tiles) stage_write(1 - buf);
   }
 
   // ---- boundary tiles (diagonal/tail) ------------------------------------
   for (int t = t_int; t < n_tiles; ++t) {
     const int buf = t & 1;
     const int kv0 = t * KBQ;
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
     __syncthreads();
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - buf);
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       const int kv0s = kv0 + sub * 32;
@@ -775,7 +777,6 @@ void attn_bwd_dq_kernel(
       }
       dsk_mfma(ds, buf, sub);
     }
-    if (t + 1 < n_tiles) stage_write(1 - buf);
   }
   __syncthreads();  // protect epilogue smem reuse
 
@@ -821,7 +822,10 @@ void attn_bwd_dkdv_kernel(
   // hardware tr16 reads from the row images. Tile height shrinks at D=128
   // so LDS (4 images) stays at 32 KB and registers below the spill line —
   // QTF=64 at D=128 measured 2x SLOWER (18.8 vs 9.0 ms) from occupancy.
-  constexpr int QTF = (D <= 64) ? 64 : 32;  // q rows per staged tile
+#ifndef ATT_QTF64
+#define ATT_QTF64 64  // q rows per staged dkdv tile at D<=64 (A/B knob)
+#endif
+  constexpr int QTF = (D <= 64) ? ATT_QTF64 : 32;  // q rows per staged tile
   constexpr int NSUB = QTF / 32;            // 32-row compute subtiles
   constexpr int IMG2 = QTF * D * 2; // bytes per [QTF][D] image
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -876,31 +880,34 @@ void attn_bwd_dkdv_kernel(
   const int n_tiles = (S + QTF - 1) / QTF;
   const int my_k_min = k0;
 
-  constexpr int NCHUNK = QTF * D / 8 / ATT_BLOCK;
-  bf16x8 q_stage[NCHUNK], do_stage[NCHUNK];
   float lse_r = 0.f, del_r = 0.f;
 
-  // SRSRC staging (cf. fwd). q and dout have separate strides (packed QKV
-  // input vs contiguous dout) -> separate descriptors/voffsets.
+  // LDS-DMA staging (cf. fwd). q and dout have separate strides (packed
+  // QKV input vs contiguous dout) -> separate descriptors/voffsets. The
+  // tiny lse/delta side-channel stays register-staged (loads at issue
+  // time, LDS writes after compute).
+  constexpr int NW = QTF * D * 2 / 1024 / WAVES;
   const auto qrs = make_rsrc(q + ibase, ((long)(S - 1) * rs_i + D) * 2);
   const auto ors = make_rsrc(dout + obase, ((long)(S - 1) * rs_o + D) * 2);
-  int voff_q[NCHUNK], voff_o[NCHUNK];
+  int dvoff_q[NW], dvoff_o[NW];
 #pragma unroll
-  for (int i = 0; i < NCHUNK; ++i) {
-    const int c = i * ATT_BLOCK + threadIdx.x;
-    const int row = c / (D / 8);
-    const int col = (c % (D / 8)) * 8;
-    voff_q[i] = (int)(((long)row * rs_i + col) * 2);
-    voff_o[i] = (int)(((long)row * rs_o + col) * 2);
+  for (int i = 0; i < NW; ++i) {
+    const int dest = (wave * NW + i) * 1024 + lane * 16;
+    const int row = dest / (D * 2);
+    const int src = (int)(dest ^ (swz_field(row & 15u, D == 64) << 4));
+    const int srow = src / (D * 2), scol = src % (D * 2);
+    dvoff_q[i] = (int)((long)srow * rs_i * 2 + scol);
+    dvoff_o[i] = (int)((long)srow * rs_o * 2 + scol);
   }
   const int tsoff_q = (int)(QTF * rs_i * 2);
   const int tsoff_o = (int)(QTF * rs_o * 2);
 
-  auto stage_load = [&](int t) {
+  auto dma_stage = [&](int t, int b) {
 #pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      q_stage[i] = rsrc_load16(qrs, voff_q[i], t * tsoff_q);
-      do_stage[i] = rsrc_load16(ors, voff_o[i], t * tsoff_o);
+    for (int i = 0; i < NW; ++i) {
+      const int win = (wave * NW + i) * 1024;
+      lds_dma16(qrs, (char*)q_img(b) + win, dvoff_q[i], t * tsoff_q);
+      lds_dma16(ors, (char*)do_img(b) + win, dvoff_o[i], t * tsoff_o);
     }
     if (threadIdx.x < QTF) {
       const long qi = (long)t * QTF + threadIdx.x;
@@ -909,23 +916,15 @@ void attn_bwd_dkdv_kernel(
       del_r = (qi < S) ? delta[bh * (long)S + qi] : 0.f;
     }
   };
-  auto stage_write = [&](int b) {
-#pragma unroll
-    for (int i = 0; i < NCHUNK; ++i) {
-      const int c = i * ATT_BLOCK + threadIdx.x;
-      const int row = c / (D / 8);
-      const int col = (c % (D / 8)) * 8;
-      lds_store16(q_img(b), row, D * 2, col * 2, q_stage[i]);
-      lds_store16(do_img(b), row, D * 2, col * 2, do_stage[i]);
-    }
+  auto misc_write = [&](int b) {
     if (threadIdx.x < QTF) {
       lse_buf[b * QTF + threadIdx.x] = lse_r;
       del_buf[b * QTF + threadIdx.x] = del_r;
     }
   };
 
-  stage_load(t0);
-  stage_write(t0 & 1);
+  dma_stage(t0, t0 & 1);
+  misc_write(t0 & 1);
 
   // Hoisted LDS read addresses (cf. fwd); q and do share the layout.
   const int pv_tj = lane & 15;
@@ -933,7 +932,7 @@ void attn_bwd_dkdv_kernel(
   unsigned qk_addr[D / 16];
 #pragma unroll
   for (int kk = 0; kk < D / 16; ++kk) {
-    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq);
+    qk_addr[kk] = swz((unsigned)(lq * (D * 2) + kk * 32 + hi * 16), lq, D == 64);
   }
   unsigned tr_addr[2][D / 32];
 #pragma unroll
@@ -942,7 +941,7 @@ void attn_bwd_dkdv_kernel(
     for (int db = 0; db < D / 32; ++db) {
       const int key0 = 8 * hi + 4 * rd + (pv_tj >> 2);
       const int dhc0 = db * 32 + 16 * pv_tg1 + 4 * (pv_tj & 3);
-      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0);
+      tr_addr[rd][db] = swz((unsigned)(key0 * (D * 2) + dhc0 * 2), key0, D == 64);
     }
   }
 
@@ -1045,22 +1044,25 @@ void attn_bwd_dkdv_kernel(
   int t_full_end = S / QTF;
   if (t_full_end > n_tiles) t_full_end = n_tiles;
   for (int t = t0; t <= t_diag_end; ++t) {
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
     __syncthreads();
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - (t & 1));
     masked_tile(t);
-    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
+    if (t + 1 < n_tiles) misc_write(1 - (t & 1));
   }
   for (int t = t_diag_end + 1; t < t_full_end; ++t) {
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
     __syncthreads();
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - (t & 1));
     interior_tile(t);
-    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
+    if (t + 1 < n_tiles) misc_write(1 - (t & 1));
   }
   for (int t = max(t_full_end, t_diag_end + 1); t < n_tiles; ++t) {
-    if (t + 1 < n_tiles) stage_load(t + 1);
+    vm_wait0();
     __syncthreads();
+    if (t + 1 < n_tiles) dma_stage(t + 1, 1 - (t & 1));
     masked_tile(t);
-    if (t + 1 < n_tiles) stage_write(1 - (t & 1));
+    if (t + 1 < n_tiles) misc_write(1 - (t & 1));
   }
   __syncthreads();  // protect epilogue smem reuse
 
