@@ -289,9 +289,34 @@ class FedServer:
                 # keys; re-key it to the new group
                 self.watchdog.rank = new_rank
                 self.watchdog.world_size = new_world
+                # REQUEUE the dead ranks' clients onto the survivors
+                # (reference node_manager_app.py:574-579: a failed worker's
+                # cid goes back on the work list). Deterministic from the
+                # agreed alive set, so every survivor runs the same split.
+                requeue = [cid for r in dead for cid in assignment[r]]
+                my_extra = [
+                    cid for i, cid in enumerate(requeue)
+                    if i % new_world == new_rank
+                ]
+                for cid in my_extra:
+                    try:
+                        local_payload, n_samples, metrics = self.client.fit(
+                            cid, outgoing, self.layout, server_round,
+                            reset_optimizer=bool(fl.get("reset_optimizer",
+                                                        True))
+                            and not self.aggregate_momenta,
+                        )
+                        local_sum.add_(local_payload, alpha=n_samples)
+                        local_weight += n_samples
+                        steps_done_max = max(
+                            steps_done_max, int(metrics.get("steps_done", 0))
+                        )
+                        dead_clients -= 1  # recovered, not a failure
+                    except Exception as e:
+                        print(f"[fed] requeued client {cid} failed: {e!r}")
 
-        # agree on failures across ranks (+ dead ranks' clients, identical
-        # on every survivor so added after the sum)
+        # agree on failures across ranks (+ dead ranks' unrecovered
+        # clients, identical on every survivor so added after the sum)
         fail_total = sum(self.comm.all_gather_scalars(float(failures)))
         fail_total += dead_clients
         if fail_total > self.accept_failures_cnt and not self.ignore_failed_rounds:

@@ -105,7 +105,8 @@ class Trainer:
             cfg = getattr(self.model, "cfg", None)
             if cfg is not None:
                 d = int(getattr(cfg, "d_model", 768))
-            table = [(1024, 32), (2048, 16), (2560, 8), (4096, 8)]
+            # mb sweep r02: 1B (d2048) mb32 91.7k > mb16 90.7k > mb8 87.7k
+            table = [(2048, 32), (2560, 8), (4096, 8)]
             for width, mb in table:
                 if d <= width:
                     return mb
