@@ -17,8 +17,8 @@ static int attn_lds_bytes(int D, int kind) {
   // Transposed A-fragments are hardware tr16 reads — no transposed images.
   const int img = 64 * D;  // bytes of one [32][D] bf16 image
   // dkdv: double-buffered [QTF][D] q + do images + 2*QTF lse/delta floats
-  // (QTF = 64 at D<=64, 32 at D=128 — mirrors the kernel's constexpr)
-  const int qtf = D <= 64 ? 64 : 32;
+  // (QTF = 64 at every D — mirrors the kernel's ATT_QTF* constexprs)
+  const int qtf = 64;
   // dq: double-buffered [64][D] k + v row images
   int imgs = kind == 0 ? 4 * (KBF * D * 2)
                        : (kind == 1 ? 4 * (64 * D * 2)

@@ -825,7 +825,13 @@ void attn_bwd_dkdv_kernel(
 #ifndef ATT_QTF64
 #define ATT_QTF64 64  // q rows per staged dkdv tile at D<=64 (A/B knob)
 #endif
-  constexpr int QTF = (D <= 64) ? ATT_QTF64 : 32;  // q rows per staged tile
+#ifndef ATT_QTF128
+// 64 q-rows per staged dkdv tile at D=128: with LDS-DMA staging the
+// registers no longer bind, and halving the barrier count measured
+// 176 -> 207 TF/s (r01's QTF=32 choice was a register-pressure artifact).
+#define ATT_QTF128 64
+#endif
+  constexpr int QTF = (D <= 64) ? ATT_QTF64 : ATT_QTF128;
   constexpr int NSUB = QTF / 32;            // 32-row compute subtiles
   constexpr int IMG2 = QTF * D * 2; // bytes per [QTF][D] image
   extern __shared__ __attribute__((aligned(16))) char smem[];

@@ -151,7 +151,7 @@ void bench_bwd(int B, int H, int S, int iters) {
   hipLaunchKernelGGL((attn_fwd_kernel<D>), grid, dim3(ATT_BLOCK), lds_fwd, 0,
                      q, k, v, slopes, o, lse, S, H, 1, bs0, hs0, rs0, bs0, hs0, rs0);
   CHECK(hipDeviceSynchronize());
-  const int qtf = D <= 64 ? 64 : 32;
+  const int qtf = 64;  // matches ATT_QTF* kernel constexprs
   const int lds_dq = 4 * 64 * D * 2 > WAVES * 64 * D ? 4 * 64 * D * 2 : WAVES * 64 * D;
   int lds_kv = 4 * qtf * D * 2 + 4 * qtf * 4;
   if (lds_kv < WAVES * 64 * D) lds_kv = WAVES * 64 * D;
