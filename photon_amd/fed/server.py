@@ -311,12 +311,15 @@ class FedServer:
                         steps_done_max = max(
                             steps_done_max, int(metrics.get("steps_done", 0))
                         )
-                        dead_clients -= 1  # recovered, not a failure
+                        # recovered: the all-gather below subtracts it from
+                        # the (globally identical) dead_clients count
+                        failures -= 1
                     except Exception as e:
                         print(f"[fed] requeued client {cid} failed: {e!r}")
 
-        # agree on failures across ranks (+ dead ranks' unrecovered
-        # clients, identical on every survivor so added after the sum)
+        # agree on failures across ranks: local failures minus local
+        # recoveries, plus the dead ranks' client count (identical on every
+        # survivor so added after the sum)
         fail_total = sum(self.comm.all_gather_scalars(float(failures)))
         fail_total += dead_clients
         if fail_total > self.accept_failures_cnt and not self.ignore_failed_rounds:

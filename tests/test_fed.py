@@ -517,3 +517,78 @@ def test_restore_cent_from_composer_pt(tiny_cfg, tmp_path):
     srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
     srv2.initialize()
     assert float(srv2.strategy.params.abs().sum()) == 0.0
+
+
+@pytest.mark.timeout(300)
+def test_watchdog_multi_survivor_agreement(tiny_cfg, tmp_path):
+    """3 ranks, rank 2 killed mid-fit: BOTH survivors must agree on the
+    alive set via the decider protocol (first atomic claim seals it),
+    rebuild to world 2, requeue the dead rank's client, and end the round
+    with identical replicated params."""
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["fl"]["n_total_clients"] = 3
+    cfg["fl"]["n_clients_per_round"] = 3
+    cfg["photon"]["checkpoint"] = False
+    cfg["photon"]["fit_timeout_s"] = 6
+    cfg["fl"]["accept_failures_cnt"] = 0
+    from tests.conftest import free_port
+
+    port = free_port()
+    flag = str(tmp_path / "rank2_fitting")
+    out_dir = str(tmp_path / "out")
+    os.makedirs(out_dir, exist_ok=True)
+    ctx = mp.get_context("spawn")
+
+    def worker(rank):
+        return ctx.Process(
+            target=_watchdog_worker_n,
+            args=(rank, 3, port, cfg, out_dir, flag, 2),
+        )
+
+    procs = [worker(r) for r in range(3)]
+    for p in procs:
+        p.start()
+    import time as _time
+
+    deadline = _time.time() + 60
+    while not os.path.exists(flag) and _time.time() < deadline:
+        _time.sleep(0.1)
+    assert os.path.exists(flag), "rank 2 never reached its fit"
+    procs[2].kill()
+    for r in (0, 1):
+        procs[r].join(timeout=180)
+        assert procs[r].exitcode == 0, f"survivor rank {r} failed"
+    procs[2].join(timeout=30)
+    o0 = torch.load(os.path.join(out_dir, "wd_0.pt"))
+    o1 = torch.load(os.path.join(out_dir, "wd_1.pt"))
+    assert o0["world"] == 2 and o1["world"] == 2
+    assert o0["failures_r1"] == 0.0, "requeue must recover the dead client"
+    assert torch.equal(o0["params"], o1["params"]), "replicas must agree"
+
+
+def _watchdog_worker_n(rank, world, port, cfg, out_dir, flag_path, hang_rank):
+    import time as _time
+
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    srv = FedServer(cfg, Comm(rank, world), "cpu")
+    if rank == hang_rank:
+        def hang(*a, **k):
+            open(flag_path, "w").close()
+            _time.sleep(600)
+
+        srv.client.fit = hang
+    srv.initialize()
+    m1 = srv.run_round(1)
+    m2 = srv.run_round(2)
+    torch.save(
+        {"params": srv.strategy.params, "world": srv.comm.world_size,
+         "failures_r1": m1["server/failures"],
+         "failures_r2": m2["server/failures"]},
+        os.path.join(out_dir, f"wd_{rank}.pt"),
+    )
+    if dist.is_initialized():
+        dist.destroy_process_group()
