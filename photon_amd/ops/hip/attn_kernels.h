@@ -1,26 +1,36 @@
 // Flash attention (causal, fused ALiBi) for CDNA4 gfx950 — MFMA bf16.
 //
 // Replaces flash-attn 2.6.3's CUDA kernels (reference install_env.sh:71;
-// SURVEY.md §2.3 rows 1-2). MI355X-native structure (not a port):
+// SURVEY.md §2.3 rows 1-2). MI355X-native structure (not a port); round-2
+// state (measured numbers in profiles/r02_final_SUMMARY.md):
 //
 //   * 256-thread workgroups = 4 wave64s; each wave owns a 32-row Q block
 //     (fwd / dQ) or 32 keys (dK/dV); mfma_f32_32x32x16_bf16 tiles.
 //   * "Swapped" QK^T — mfma(A=K, B=Q) gives S[key][q] with q = lane&31, so
 //     the online-softmax state (m, l) is lane-local: rescales are scalar
 //     per lane, row reductions are 16 regs + one shfl_xor(32).
+//   * exp2-DOMAIN softmax: scale/slopes pre-multiplied by log2(e), every
+//     exponential is one native v_exp_f32; LSE converted to natural log at
+//     the epilogue (wire format unchanged). Defer-max (T13, THR=8).
+//   * The KV loop is SPLIT into interior tiles (every element unmasked for
+//     the whole block: straight-line softmax, zero compares) and boundary
+//     tiles (diagonal/tail masking via the -1e30 sentinel).
+//   * K/V tiles reach LDS by LDS-DMA (buffer_load..lds, 1 KiB per
+//     wave-instruction, SRSRC descriptor with flags 0x27FAC and
+//     num_records clamping rows >= S to hardware zeros): no staging
+//     registers, no ds_write issue cost. The zero-bank-conflict XOR
+//     swizzle (GF(2)-searched, row-preserving) is applied to the DMA
+//     SOURCE offsets; double-buffered, one barrier + one counted
+//     vmcnt(0) per tile. Occupancy: fwd64 4 waves/SIMD, dq64 3, others 2.
 //   * P (f32 regs) is converted to the next MFMA's B-operand fragments
-//     in-register with v_cvt_pk_bf16_f32 + permlane32_swap — no LDS
-//     round-trip for P.
-//   * K/V tiles staged in LDS: row image [32][D] with a ((row&15)<<4) XOR
-//     byte swizzle (conflict-free b128 column reads) + an explicitly
-//     transposed image [D][32] for the A-operands of PV / dQ / dK / dV
-//     (upgrade path: ds_read_b64_tr_b16 hardware transpose reads).
+//     in-register with v_cvt_pk_bf16_f32 + permlane32_swap; A-operands of
+//     PV / dQ / dK / dV come from ds_read_b64_tr_b16 hardware transpose
+//     reads of the row images — no transposed image, no LDS round-trip
+//     for P.
 //   * Online softmax carries LSE out for the backward; backward is the
 //     standard FlashAttention-2 split: one kernel for dK/dV (blocks own key
 //     tiles, loop over Q) and one for dQ (blocks own Q tiles, loop over
 //     KV) — atomics-free and bit-deterministic.
-//
-// v1 is correctness-first: single-buffered LDS, one barrier pair per tile.
 
 #pragma once
 #include "common.h"
