@@ -111,29 +111,6 @@ DEV_INLINE bf16x8 pack_bfrag(const float* p, int rb) {
   return out.v;
 }
 
-// Load a 16-byte (8 x bf16) fragment from an LDS row image with the XOR
-// swizzle. byte = row*rowstride + coloff must be 16B-aligned pre-swizzle.
-DEV_INLINE bf16x8 lds_frag(const __bf16* img, int row, int rowstride_b,
-                           int coloff_b) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
-                      rowstride_b == 128);
-  return *(const bf16x8*)((const char*)img + byte);
-}
-
-DEV_INLINE void lds_store16(__bf16* img, int row, int rowstride_b,
-                            int coloff_b, bf16x8 v) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
-                      rowstride_b == 128);
-  *(bf16x8*)((char*)img + byte) = v;
-}
-
-DEV_INLINE void lds_store2(__bf16* img, int row, int rowstride_b,
-                           int coloff_b, __bf16 v) {
-  unsigned byte = swz((unsigned)(row * rowstride_b + coloff_b), row,
-                      rowstride_b == 128);
-  *(__bf16*)((char*)img + byte) = v;
-}
-
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 constexpr float LOG2E = 1.4426950408889634f;
@@ -170,13 +147,6 @@ DEV_INLINE __amdgpu_buffer_rsrc_t make_rsrc(const __bf16* base,
                                            (int)num_bytes, 0x27FAC);
 }
 
-DEV_INLINE bf16x8 rsrc_load16(__amdgpu_buffer_rsrc_t rsrc, int voffset,
-                              int soffset) {
-  union { f32x4 f; bf16x8 b; } u;
-  u.f = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voffset, soffset, 0);
-  return u.b;
-}
-
 // LDS-DMA: one wave-instruction moves 1 KiB (16 B/lane) global -> LDS with
 // NO data registers and no ds_write issue cost (buffer_load_dwordx4 ..lds).
 // The LDS destination is linear in lanes, so the ROW-PRESERVING XOR
@@ -207,37 +177,6 @@ DEV_INLINE bf16x4 lds_tr16(const __bf16* img, unsigned byte) {
   typedef __attribute__((address_space(3))) char* lds_cp;
   return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
       (lds_v4p)((lds_cp)img + byte));
-}
-
-// Stage a [rows=32][D] global tile into (a) the row image (swizzled,
-// row stride D*2 bytes) and (b) the transposed image [D][32] (swizzled,
-// row stride 64 bytes). Cooperative across the whole 256-thread block.
-// Rows >= rows_valid are zero-filled.
-template <int D>
-DEV_INLINE void stage_tile(const __bf16* __restrict__ gsrc, long g_row0,
-                           long g_rows_total, long g_row_stride,
-                           __bf16* row_img, __bf16* t_img) {
-  constexpr int CHUNKS = 32 * D / 8;  // 16B chunks
-  for (int c = threadIdx.x; c < CHUNKS; c += ATT_BLOCK) {
-    const int row = c / (D / 8);
-    const int col = (c % (D / 8)) * 8;
-    bf16x8 v;
-    const long grow = g_row0 + row;
-    if (grow < g_rows_total) {
-      v = *(const bf16x8*)(gsrc + grow * g_row_stride + col);
-    } else {
-      v = bf16x8{};
-    }
-    if (row_img) lds_store16(row_img, row, D * 2, col * 2, v);
-#ifndef ABENCH_NO_VT
-    if (t_img) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        lds_store2(t_img, col + j, 64, row * 2, v[j]);
-      }
-    }
-#endif
-  }
 }
 
 // ---------------------------------------------------------------------------
