@@ -101,3 +101,21 @@ def test_duration_parse_property(n):
     assert duration_to_batches(f"{n}ba") == n
     assert duration_to_batches(n) == n
     assert duration_to_batches(str(n)) == n
+
+
+def test_v03_presets_compose():
+    """The vendored reference tasks_v0.3 / eval_gauntlet_v0.3 groups load
+    through the hydra-surface engine (the reference's exact preset names)."""
+    from photon_amd.conf import compose, config_yaml_dir
+
+    cfg = compose(config_yaml_dir(), "base",
+                  ["icl_tasks_config=tasks_v0.3",
+                   "eval_gauntlet_config=eval_gauntlet_v0.3"])
+    tasks = cfg["icl_tasks_config"]["icl_tasks"]
+    assert len(tasks) == 32
+    kinds = {str(t["icl_task_type"]) for t in tasks}
+    assert kinds == {"language_modeling", "multiple_choice", "schema",
+                     "generation_task_with_answers"}
+    g = cfg["eval_gauntlet_config"]["eval_gauntlet"]
+    assert len(g["categories"]) == 5
+    assert "core_average" in g["averages"]

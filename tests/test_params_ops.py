@@ -300,3 +300,40 @@ def test_fed_round_momenta_plus_personalization(tmp_path):
     assert m1["server/sampled_clients"] == 2
     assert float(srv.client_m1.abs().sum()) > 0
     assert torch.isfinite(srv.strategy.params).all()
+
+
+def test_dma_swizzle_source_permutation():
+    """The LDS-DMA source-offset formula must be a bijection of each 1 KiB
+    window whose inverse is the row-local XOR swizzle — i.e. for every
+    destination chunk, reading global byte dest^field(row) and landing it
+    linearly reproduces exactly the swizzled image the compute-side reads
+    expect (mirrors attn_kernels.h dma_voff + swz/swz_field)."""
+
+    def swz_field(r, d64):
+        r &= 15
+        if d64:
+            return (((r >> 1) & 1) << 2) | ((r >> 2) & 1) | (((r >> 3) & 1) << 1)
+        return ((r & 3) << 2) | (r >> 2)
+
+    for D in (64, 128):
+        rowstride = D * 2
+        d64 = D == 64
+        img_bytes = 64 * rowstride  # one staged [64][D] bf16 tile
+        # forward map used by the COMPUTE side: byte -> byte ^ field(row)<<4
+        def swz(b):
+            return b ^ (swz_field((b // rowstride) & 15, d64) << 4)
+
+        # DMA side: dest chunk (16B) at d reads source chunk s = d^field
+        for win in range(0, img_bytes, 1024):
+            seen = set()
+            for lane_byte in range(win, win + 1024, 16):
+                row = lane_byte // rowstride
+                src = lane_byte ^ (swz_field(row & 15, d64) << 4)
+                # row-preserving: source stays in the same row
+                assert src // rowstride == row, (D, lane_byte)
+                # the value landing at `lane_byte` must be what the
+                # compute-side read of unswizzled byte `src` expects:
+                assert swz(src) == lane_byte, (D, lane_byte)
+                seen.add(src)
+            # bijection within the window
+            assert len(seen) == 64 and all(win <= s < win + 1024 for s in seen)
