@@ -38,7 +38,12 @@ from . import hip_ext
 #       BGRADB disables split-K on the reduction-heavy dW GEMMs, MPT-1B
 #       step 1704 -> 3515 ms; kept for re-evaluation on newer hipblaslt);
 #   "off"           — plain torch linears.
+import functools
+
+
+@functools.lru_cache(maxsize=1)
 def _lt_mode() -> str:
+    # cached: read once per process (384 linear calls per 125M step)
     return os.environ.get("PHOTON_LT_MODE", "cpp")
 
 _CUSTOM_FWD = torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
@@ -47,12 +52,13 @@ _CUSTOM_BWD = torch.amp.custom_bwd(device_type="cuda")
 
 def lt_available(x: torch.Tensor) -> bool:
     ext = hip_ext()
-    return (
-        ext is not None
-        and hasattr(ext, "lt_linear_fwd")
-        and x.is_cuda
-        and _lt_mode() != "off"
-    )
+    if ext is None or not x.is_cuda:
+        return False
+    mode = _lt_mode()
+    if mode == "off":
+        return False
+    need = "fused_linear" if mode == "cpp" else "lt_linear_fwd"
+    return hasattr(ext, need)
 
 
 class _LtLinearFn(torch.autograd.Function):
