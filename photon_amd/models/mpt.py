@@ -84,9 +84,16 @@ class MPTAttention(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, D = x.shape
         H, dh = self.n_heads, self.d_head
-        # hipBLASLt epilogue path: bias fused into the GEMM, bias-grad
-        # fused into the backward dW GEMM (BGRADB) — no torch bias kernels
-        fused = lt_available(x)
+        # Fused-linear path (C++ autograd + HIP bias-grad backward). Only
+        # for PLAIN nn.Linear projections: tensor-parallel swaps replace
+        # them with Column/RowParallelLinear whose forward carries the TP
+        # collectives — bypassing those with raw .weight would silently
+        # drop the all-reduces.
+        fused = (
+            lt_available(x)
+            and type(self.Wqkv) is nn.Linear
+            and type(self.out_proj) is nn.Linear
+        )
         qkv = (
             lt_linear(x, self.Wqkv.weight, self.Wqkv.bias)
             if fused
@@ -115,8 +122,9 @@ class MPTMLP(nn.Module):
         self.down_proj = nn.Linear(hidden, cfg.d_model, bias=bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if lt_available(x):
-            # one autograd node: GELU_AUX_BIAS fwd, DGELU_BGRAD/BGRADB bwd
+        # plain-Linear guard: see MPTAttention.forward (TP swap safety)
+        if (lt_available(x) and type(self.up_proj) is nn.Linear
+                and type(self.down_proj) is nn.Linear):
             return lt_mlp(x, self.up_proj.weight, self.up_proj.bias,
                           self.down_proj.weight, self.down_proj.bias)
         return self.down_proj(self.act(self.up_proj(x)))

@@ -140,3 +140,24 @@ def test_ddp_two_rank_sync(tmp_path):
     # post-sync grads identical on both ranks and equal to the mean of pre
     assert torch.allclose(a["post"], b["post"], atol=1e-7)
     assert torch.allclose(a["post"], (a["pre"] + b["pre"]) / 2, atol=1e-6)
+
+
+def test_tp_swap_disables_fused_linear_path(tiny_cfg):
+    """TP replaces the projection Linears; the fused lt_linear path must
+    detect the swap and fall back to module forwards (otherwise the TP
+    collectives would be silently skipped on GPU)."""
+    import torch.nn as nn
+
+    from photon_amd.models import build_model
+    from photon_amd.parallel.tp import apply_tensor_parallel
+
+    m = build_model(tiny_cfg["llm_config"])
+    apply_tensor_parallel(m, rank=0, world=2)  # world=1 is a no-op
+    blk = m.transformer.blocks[0]
+    assert type(blk.attn.Wqkv) is not nn.Linear
+    assert type(blk.ffn.up_proj) is not nn.Linear
+    # forward must route through the swapped modules (shape check only:
+    # without an initialized group the row-parallel all-reduce is local)
+    ids = __import__("torch").randint(0, 512, (1, 32))
+    out = m(ids)
+    assert out["logits"].shape[-1] == tiny_cfg["llm_config"]["model"]["vocab_size"]
