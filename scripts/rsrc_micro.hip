@@ -1,4 +1,8 @@
 #include <hip/hip_runtime.h>
+// flags=0 (default) REPRODUCES the silent-zeros bug: the V# DATA_FORMAT
+// field is invalid and every load is treated as out-of-range. Build with
+// -DRSRC_FLAGS=0x27FAC (dst_sel XYZW | NFMT float | DFMT 32) for correct
+// loads — the constant attn_kernels.h::make_rsrc ships with.
 #ifndef RSRC_FLAGS
 #define RSRC_FLAGS 0
 #endif

@@ -1,5 +1,14 @@
 """Search 16-entry XOR tables (byte ^= T[row&15]<<4) minimizing LDS bank
-conflicts across the three access patterns of attn_kernels.h."""
+conflicts across the three access patterns of attn_kernels.h
+(b128 column reads, ds_read_b64_tr_b16 transpose reads, b128 stores).
+
+SHIPPED tables (attn_kernels.h swz_field):
+ - D=128: 4-bit nibble half-swap T[r] = ((r&3)<<2)|(r>>2) — zero conflicts.
+ - D=64:  the LDS-DMA staging needs a ROW-PRESERVING field (<=7, rows are
+   128 B), so the shipped table is the best 3-bit GF(2)-linear map
+   (columns m=(0,4,1,2)): also zero conflicts for all three patterns.
+Re-run with the 3-bit constraint (see the round-2 session) if the access
+patterns change."""
 import itertools
 
 BANKS = 64
