@@ -147,6 +147,9 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
         nf = task.get("num_fewshot", [0])
         n_fewshot = int(nf[0]) if isinstance(nf, (list, tuple)) else int(nf)
         do_norm = bool(task.get("do_normalization", True))
+        has_categories = bool(task.get("has_categories", False))
+        cat_correct: dict[str, float] = {}
+        cat_total: dict[str, float] = {}
         try:
             examples = load_jsonl_task(task["dataset_uri"])
         except OSError:
@@ -158,6 +161,7 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
         for i, ex in enumerate(examples):
             shots = _fewshot_prefix(examples, i, n_fewshot, kind, delim,
                                     prelim)
+            ex_correct, ex_total = 0, 0
             if kind == "language_modeling":
                 ce, n, c = _continuation_stats(
                     model, tokenizer, shots + ex["context"] + delim,
@@ -165,6 +169,7 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                 )
                 correct += c
                 total += n
+                ex_correct, ex_total = c, n
             elif kind == "multiple_choice":
                 ces = []
                 for choice in ex["choices"]:
@@ -173,9 +178,11 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                         choice, device, max_seq_len,
                     )
                     ces.append(ce / max(n, 1))
-                correct += int(min(range(len(ces)), key=ces.__getitem__)
-                               == int(ex["gold"]))
+                ok_mc = int(min(range(len(ces)), key=ces.__getitem__)
+                            == int(ex["gold"]))
+                correct += ok_mc
                 total += 1
+                ex_correct, ex_total = ok_mc, 1
             elif kind == "schema":
                 # choose the context option that best explains the SAME
                 # continuation (winograd-style)
@@ -186,9 +193,11 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                         ex["continuation"], device, max_seq_len,
                     )
                     ces.append(ce / max(n, 1))
-                correct += int(min(range(len(ces)), key=ces.__getitem__)
-                               == int(ex["gold"]))
+                ok_sc = int(min(range(len(ces)), key=ces.__getitem__)
+                            == int(ex["gold"]))
+                correct += ok_sc
                 total += 1
+                ex_correct, ex_total = ok_sc, 1
             elif kind == "generation_task_with_answers":
                 gen = _greedy_generate(
                     model, tokenizer, shots + prelim + ex["context"] + delim,
@@ -206,11 +215,22 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                     ok = any(a.strip() == gen.strip() for a in answers)
                 correct += int(ok)
                 total += 1
+                ex_correct, ex_total = int(ok), 1
             else:
                 raise ValueError(f"unknown icl_task_type {kind!r}")
+            if has_categories and "category" in ex:
+                c_ = str(ex["category"])
+                cat_correct[c_] = cat_correct.get(c_, 0) + ex_correct
+                cat_total[c_] = cat_total.get(c_, 0) + ex_total
         results[f"metrics/icl/{label}/accuracy"] = (
             correct / total if total else float("nan")
         )
+        # per-category breakdown (jeopardy-style has_categories tasks)
+        for c_, tot in cat_total.items():
+            if tot:
+                results[f"metrics/icl/{label}/{c_}/accuracy"] = (
+                    cat_correct[c_] / tot
+                )
     model.train()
     return results
 

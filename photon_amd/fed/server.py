@@ -109,6 +109,10 @@ class FedServer:
         )
         self.server_steps_cumulative = 0
         self.start_round = 1
+        # cumulative wall time across resumes (reference state.bin
+        # time_offset, s3_utils.py:374-389)
+        self.time_offset = 0.0
+        self._t_started = time.time()
         # aggregated client momenta when fl.aggregate_momenta
         self.aggregate_momenta = bool(fl.get("aggregate_momenta", False))
         self.client_m1 = self.layout.like() if self.aggregate_momenta else None
@@ -149,6 +153,7 @@ class FedServer:
                 self.saving_path, self.run_uuid, target, self.strategy, self.layout
             )
             self.server_steps_cumulative = int(state.get("server_steps_cumulative", 0))
+            self.time_offset = float(state.get("time_offset", 0.0))
             self.history.load_state(state.get("history", {}))
             # client_state round-trips as str() in state.bin (reference
             # s3_utils.py:374-389); restore steps_done per client so lr
@@ -470,6 +475,7 @@ class FedServer:
                     self.history.state(),
                     {cid: vars(st) for cid, st in self.client.client_states.items()},
                     self.server_steps_cumulative,
+                    time_offset=self.time_offset + (time.time() - self._t_started),
                     client_momenta=(
                         (self.client_m1, self.client_m2)
                         if self.aggregate_momenta

@@ -592,3 +592,38 @@ def _watchdog_worker_n(rank, world, port, cfg, out_dir, flag_path, hang_rank):
     )
     if dist.is_initialized():
         dist.destroy_process_group()
+
+
+def test_time_offset_accumulates_across_resume(tiny_cfg, tmp_path):
+    """state.bin time_offset carries cumulative wall time across resumes
+    (reference s3_utils.py:374-389)."""
+    import pickle
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg["photon"]["checkpoint"] = True
+    cfg["photon"]["saving_path"] = str(tmp_path)
+    srv = FedServer(cfg, Comm(0, 1), "cpu")
+    srv.run(2)
+    with open(tmp_path / cfg["run_uuid"] / "server" / "2" / "state.bin",
+              "rb") as f:
+        t1 = pickle.load(f)["time_offset"]
+    assert t1 > 0
+
+    cfg2 = copy.deepcopy(cfg)
+    cfg2["photon"]["resume_round"] = -1
+    srv2 = FedServer(cfg2, Comm(0, 1), "cpu")
+    srv2.initialize()
+    assert srv2.time_offset == t1
+    srv2.run_round(3)
+    # force a checkpoint of round 3 via the run loop bookkeeping
+    from photon_amd.fed.server_ckpt import upload_server_checkpoint
+
+    upload_server_checkpoint(
+        srv2.saving_path, srv2.run_uuid, 3, srv2.strategy, srv2.layout,
+        srv2.history.state(), {}, srv2.server_steps_cumulative,
+        time_offset=srv2.time_offset + 1.0,
+    )
+    with open(tmp_path / cfg["run_uuid"] / "server" / "3" / "state.bin",
+              "rb") as f:
+        t2 = pickle.load(f)["time_offset"]
+    assert t2 > t1

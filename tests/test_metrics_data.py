@@ -304,3 +304,47 @@ def test_mc4_split_table():
         split_spec("xx", "train")
     with pytest.raises(ValueError):
         split_spec("de", "val_xsmall")  # non-en has no truncated variants
+
+
+def test_icl_per_category_reporting():
+    """has_categories tasks (jeopardy-style) report per-category accuracy
+    alongside the overall (reference llm-foundry category breakdown)."""
+    import json
+
+    import torch
+
+    from photon_amd.data.convert import load_tokenizer
+    from photon_amd.eval import evaluate_icl_tasks
+    from photon_amd.models import build_model
+
+    import tempfile
+    with tempfile.TemporaryDirectory() as d:
+        path = f"{d}/cat_task.jsonl"
+        with open(path, "w") as f:
+            for i, cat in enumerate(["history", "science", "history"]):
+                f.write(json.dumps({
+                    "context": f"Q{i}: the answer is",
+                    "continuation": " yes",
+                    "category": cat,
+                }) + "\n")
+        torch.manual_seed(3)
+        model = build_model({
+            "model": {"d_model": 64, "n_heads": 2, "n_layers": 1,
+                      "expansion_ratio": 2, "max_seq_len": 128,
+                      "vocab_size": 512,
+                      "attn_config": {"attn_impl": "torch"}}
+        })
+        tok = load_tokenizer(None)
+        res = evaluate_icl_tasks(
+            model,
+            [{"label": "jeo", "dataset_uri": path,
+              "icl_task_type": "language_modeling",
+              "has_categories": True}],
+            tok, max_seq_len=128,
+        )
+    assert "metrics/icl/jeo/accuracy" in res
+    assert "metrics/icl/jeo/history/accuracy" in res
+    assert "metrics/icl/jeo/science/accuracy" in res
+
+
+
