@@ -145,11 +145,10 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
         cot = task.get("cot_delimiter")
         stops = list(task.get("early_stopping_criteria") or [])
         nf = task.get("num_fewshot", [0])
-        n_fewshot = int(nf[0]) if isinstance(nf, (list, tuple)) else int(nf)
+        fewshots = [int(x) for x in nf] if isinstance(nf, (list, tuple)) \
+            else [int(nf)]
         do_norm = bool(task.get("do_normalization", True))
         has_categories = bool(task.get("has_categories", False))
-        cat_correct: dict[str, float] = {}
-        cat_total: dict[str, float] = {}
         try:
             examples = load_jsonl_task(task["dataset_uri"])
         except OSError:
@@ -157,6 +156,30 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
             continue
         if limit_examples:
             examples = examples[:limit_examples]
+        # the reference evaluates EVERY num_fewshot value (e.g.
+        # arc_challenge [3, 25]); the un-suffixed key carries the first
+        for shot_i, n_fewshot in enumerate(fewshots):
+            acc, cat_acc = _eval_task_once(
+                model, tokenizer, examples, kind, delim, prelim, cot, stops,
+                n_fewshot, do_norm, has_categories, device, max_seq_len,
+            )
+            if shot_i == 0:
+                results[f"metrics/icl/{label}/accuracy"] = acc
+                for c_, a_ in cat_acc.items():
+                    results[f"metrics/icl/{label}/{c_}/accuracy"] = a_
+            if len(fewshots) > 1:
+                results[f"metrics/icl/{label}/{n_fewshot}-shot/accuracy"] = acc
+    model.train()
+    return results
+
+
+@torch.no_grad()
+def _eval_task_once(model, tokenizer, examples, kind, delim, prelim, cot,
+                    stops, n_fewshot, do_norm, has_categories, device,
+                    max_seq_len):
+    cat_correct: dict[str, float] = {}
+    cat_total: dict[str, float] = {}
+    if True:  # (kept indentation of the original loop body)
         correct, total = 0, 0
         for i, ex in enumerate(examples):
             shots = _fewshot_prefix(examples, i, n_fewshot, kind, delim,
@@ -222,17 +245,10 @@ def evaluate_icl_tasks(model, tasks_cfg, tokenizer, device="cpu",
                 c_ = str(ex["category"])
                 cat_correct[c_] = cat_correct.get(c_, 0) + ex_correct
                 cat_total[c_] = cat_total.get(c_, 0) + ex_total
-        results[f"metrics/icl/{label}/accuracy"] = (
-            correct / total if total else float("nan")
-        )
-        # per-category breakdown (jeopardy-style has_categories tasks)
-        for c_, tot in cat_total.items():
-            if tot:
-                results[f"metrics/icl/{label}/{c_}/accuracy"] = (
-                    cat_correct[c_] / tot
-                )
-    model.train()
-    return results
+    acc = correct / total if total else float("nan")
+    cat_acc = {c_: cat_correct[c_] / tot for c_, tot in cat_total.items()
+               if tot}
+    return acc, cat_acc
 
 
 def gauntlet_composite(task_results: dict[str, float],

@@ -234,10 +234,14 @@ def test_gauntlet_v03_end_to_end():
     tok = load_tokenizer(None)  # byte tokenizer
     results = evaluate_icl_tasks(model, tasks_cfg["icl_tasks"], tok,
                                  max_seq_len=256, limit_examples=2)
-    # every one of the 32 tasks produced a (finite) score on the stub data
-    scores = [v for k, v in results.items() if k.startswith("metrics/icl/")]
-    assert len(scores) == 32
-    assert all(not math.isnan(v) for v in scores), results
+    # every one of the 32 tasks produced a (finite) base score on the stub
+    # data; tasks with num_fewshot lists additionally report per-shot keys
+    base = {k: v for k, v in results.items()
+            if k.startswith("metrics/icl/") and "-shot/" not in k}
+    assert len(base) == 32
+    assert all(not math.isnan(v) for v in base.values()), results
+    shot_keys = [k for k in results if "-shot/" in k]
+    assert shot_keys, "multi-fewshot tasks must report per-shot accuracies"
     comp = gauntlet_composite(results, gauntlet_cfg)
     assert "metrics/eval_gauntlet/average" in comp
     assert "metrics/eval_gauntlet/core_average" in comp
