@@ -36,6 +36,30 @@ def make_grad_sync_hook(comm: Comm):
     return BucketedGradSync(world_size=comm.world_size)
 
 
+def run_split_eval(cfg, trainer) -> dict:
+    """centralized.split_eval (reference centralised_train.py:74): evaluate
+    every client stream separately and report per-stream CE alongside the
+    concatenated-stream eval."""
+    from .data import build_eval_loader
+
+    streams = (cfg.get("dataset", {}).get("val", {}) or {}).get("streams") \
+        or (cfg.get("dataset", {}).get("train", {}) or {}).get("streams") \
+        or []
+    out = {}
+    saved = trainer.eval_loader
+    subset = int(cfg["llm_config"].get("eval_subset_num_batches", -1))
+    try:
+        for cid in range(len(streams)):
+            trainer.eval_loader = build_eval_loader(cfg, client_id=cid)
+            m = trainer.eval(subset)
+            ce = m.get("metrics/eval/LanguageCrossEntropy")
+            if ce is not None:
+                out[f"metrics/eval/LanguageCrossEntropy_stream_{cid}"] = ce
+    finally:
+        trainer.eval_loader = saved
+    return out
+
+
 def run_icl_eval(cfg, model, device) -> dict:
     """Config-gated ICL + gauntlet evaluation (icl_tasks_config /
     eval_gauntlet_config groups, both `empty` by default)."""
@@ -135,6 +159,8 @@ def main(argv: list[str] | None = None):
 
     if cent.get("eval_only", False):
         metrics = trainer.eval()
+        if cent.get("split_eval", False):
+            metrics.update(run_split_eval(cfg, trainer))
         metrics.update(run_icl_eval(cfg, model, device))
         if rank == 0:
             history.add_metrics_centralized(0, metrics)
@@ -147,6 +173,8 @@ def main(argv: list[str] | None = None):
     duration = llm.get("max_duration", "100ba")
     metrics = trainer.fit(duration)
     eval_metrics = trainer.eval(int(llm.get("eval_subset_num_batches", -1)))
+    if cent.get("split_eval", False):
+        eval_metrics.update(run_split_eval(cfg, trainer))
     eval_metrics.update(run_icl_eval(cfg, model, device))
     if rank == 0:
         history.add_metrics_centralized(trainer.timestamp.batch, {**metrics, **eval_metrics})

@@ -240,3 +240,35 @@ def test_optimizer_bucket_with_master_only_state():
     opt.step()  # must not raise
     st = opt.state[p]
     assert "exp_avg" in st and st["step"] == 1
+
+
+def test_centralized_split_eval(tiny_cfg, tmp_path, monkeypatch):
+    """centralized.split_eval reports a per-stream CE for every configured
+    stream (reference centralised_train.py:74 surface)."""
+    import copy
+
+    from photon_amd import centralised_train as ct
+
+    cfg = copy.deepcopy(tiny_cfg)
+    cfg.setdefault("centralized", {})
+    cfg["centralized"]["split_eval"] = True
+    cfg["dataset"]["val"] = {
+        "root_local": "", "split": "validation",
+        "streams": [
+            {"client_streams": {f"c{i}": {"local": f"client_{i}"}}}
+            for i in range(3)
+        ],
+    }
+    from photon_amd.data import build_eval_loader, build_train_loader
+    from photon_amd.models import build_model
+    from photon_amd.train import Trainer
+
+    model = build_model(cfg["llm_config"])
+    tr = Trainer(model, cfg["llm_config"],
+                 train_loader=build_train_loader(cfg, client_id=0),
+                 eval_loader=build_eval_loader(cfg, client_id=0),
+                 device="cpu")
+    out = ct.run_split_eval(cfg, tr)
+    assert set(out) == {
+        f"metrics/eval/LanguageCrossEntropy_stream_{i}" for i in range(3)
+    }
